@@ -1,0 +1,330 @@
+"""ctypes bridge to the in-tree HIP kernel library (cilfw/_hip_lib.so).
+
+Raises ImportError at import time when the library isn't built — ``ops._backend``
+then refuses to run GPU compute (no silent ATen fallback).
+
+Every function allocates outputs with torch (same allocator/stream semantics as
+the rest of the program) and launches the hand-written gfx950 kernels on the
+CURRENT torch HIP stream, so ordering with surrounding torch ops is implicit and
+hipGraph capture of a training step captures these launches too.
+"""
+
+import ctypes
+import os
+
+import torch
+
+_LIB_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "_hip_lib.so")
+if not os.path.exists(_LIB_PATH):
+    raise ImportError(f"cilfw HIP library not built: {_LIB_PATH}")
+_lib = ctypes.CDLL(_LIB_PATH)
+
+_DEBUG = os.environ.get("CILFW_SYNC_DEBUG") == "1"
+
+c_vp = ctypes.c_void_p
+c_i = ctypes.c_int
+c_l = ctypes.c_long
+c_f = ctypes.c_float
+
+_lib.cilfw_sync.restype = c_i
+_lib.cilfw_error_string.restype = ctypes.c_char_p
+_lib.cilfw_error_string.argtypes = [c_i]
+
+
+def _stream():
+    return c_vp(torch.cuda.current_stream().cuda_stream)
+
+
+def _ptr(t):
+    return c_vp(0 if t is None else t.data_ptr())
+
+
+def _check(name):
+    if _DEBUG:
+        e = _lib.cilfw_sync()
+        if e != 0:
+            raise RuntimeError(
+                f"HIP error after {name}: "
+                f"{_lib.cilfw_error_string(e).decode()}")
+
+
+def _bf16(t, name):
+    assert t.dtype == torch.bfloat16, \
+        f"{name}: GPU compute path is bf16 (got {t.dtype}); run with " \
+        f"--dtype bf16 or set CILFW_FORCE_TORCH=1 for debugging"
+    assert t.is_contiguous(), f"{name}: needs contiguous tensor"
+    return t
+
+
+# ------------------------------------------------------------------------ conv
+
+def conv2d_fwd(x, w, stride, pad):
+    _bf16(x, "conv2d_fwd.x")
+    _bf16(w, "conv2d_fwd.w")
+    N, H, W_, C = x.shape
+    R, S, Cw, K = w.shape
+    assert Cw == C
+    Ho = (H + 2 * pad - R) // stride + 1
+    Wo = (W_ + 2 * pad - S) // stride + 1
+    y = torch.empty(N, Ho, Wo, K, dtype=torch.bfloat16, device=x.device)
+    _lib.cilfw_conv2d_fwd(_ptr(x), _ptr(w), _ptr(y), c_i(N), c_i(H), c_i(W_),
+                          c_i(C), c_i(K), c_i(R), c_i(S), c_i(stride),
+                          c_i(pad), c_i(Ho), c_i(Wo), _stream())
+    _check("conv2d_fwd")
+    return y
+
+
+def conv2d_bwd_data(dy, w, stride, pad, H, W_):
+    _bf16(dy, "conv2d_bwd_data.dy")
+    N, Ho, Wo, K = dy.shape
+    R, S, C, Kw = w.shape
+    assert Kw == K
+    dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
+    _lib.cilfw_conv2d_bwd_data(_ptr(dy), _ptr(w), _ptr(dx), c_i(N), c_i(H),
+                               c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
+                               c_i(stride), c_i(pad), c_i(Ho), c_i(Wo),
+                               _stream())
+    _check("conv2d_bwd_data")
+    return dx
+
+
+def conv2d_bwd_weight(dy, x, stride, pad, R, S):
+    _bf16(dy, "conv2d_bwd_weight.dy")
+    _bf16(x, "conv2d_bwd_weight.x")
+    N, H, W_, C = x.shape
+    _, Ho, Wo, K = dy.shape
+    dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
+    _lib.cilfw_conv2d_bwd_weight(_ptr(dy), _ptr(x), _ptr(dw), c_i(N), c_i(H),
+                                 c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
+                                 c_i(stride), c_i(pad), c_i(Ho), c_i(Wo),
+                                 _stream())
+    _check("conv2d_bwd_weight")
+    return dw
+
+
+# -------------------------------------------------------------------------- bn
+
+def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
+           relu):
+    _bf16(x, "bn_fwd.x")
+    C = x.shape[-1]
+    M = x.numel() // C
+    y = torch.empty_like(x)
+    mean = torch.empty(C, dtype=torch.float32, device=x.device)
+    invstd = torch.empty(C, dtype=torch.float32, device=x.device)
+    scratch = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+    gf = gamma.float().contiguous()
+    bf = beta.float().contiguous()
+    _lib.cilfw_bn_fwd(_ptr(x), _ptr(y), _ptr(gf), _ptr(bf),
+                      _ptr(running_mean), _ptr(running_var), _ptr(mean),
+                      _ptr(invstd), _ptr(scratch), c_l(M), c_i(C),
+                      c_f(momentum), c_f(eps), c_i(1 if training else 0),
+                      c_i(1 if relu else 0), _stream())
+    _check("bn_fwd")
+    return y, mean, invstd
+
+
+def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training):
+    _bf16(dy, "bn_bwd.dy")
+    C = x.shape[-1]
+    M = x.numel() // C
+    dx = torch.empty_like(x)
+    dgamma = torch.empty(C, dtype=torch.float32, device=x.device)
+    dbeta = torch.empty(C, dtype=torch.float32, device=x.device)
+    gf = gamma.float().contiguous()
+    _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(gf),
+                      _ptr(mean), _ptr(invstd), _ptr(dgamma), _ptr(dbeta),
+                      c_l(M), c_i(C), c_i(1 if relu else 0),
+                      c_i(1 if training else 0), _stream())
+    _check("bn_bwd")
+    return dx, dgamma, dbeta
+
+
+# ------------------------------------------------------------------ elementwise
+
+def add_relu_fwd(a, b):
+    _bf16(a, "add_relu.a")
+    y = torch.empty_like(a)
+    _lib.cilfw_add_relu_fwd(_ptr(a), _ptr(b), _ptr(y), c_l(a.numel()),
+                            _stream())
+    _check("add_relu_fwd")
+    return y
+
+
+def add_relu_bwd(dy, y):
+    da = torch.empty_like(dy)
+    _lib.cilfw_add_relu_bwd(_ptr(dy), _ptr(y), _ptr(da), c_l(dy.numel()),
+                            _stream())
+    _check("add_relu_bwd")
+    return da
+
+
+def downsample_a_fwd(x):
+    _bf16(x, "downsample_a.x")
+    N, H, W_, C = x.shape
+    y = torch.empty(N, H // 2, W_ // 2, 2 * C, dtype=x.dtype, device=x.device)
+    _lib.cilfw_downsample_a_fwd(_ptr(x), _ptr(y), c_i(N), c_i(H), c_i(W_),
+                                c_i(C), _stream())
+    _check("downsample_a_fwd")
+    return y
+
+
+def downsample_a_bwd(dy, H, W_):
+    N = dy.shape[0]
+    C = dy.shape[3] // 2
+    dx = torch.empty(N, H, W_, C, dtype=dy.dtype, device=dy.device)
+    _lib.cilfw_downsample_a_bwd(_ptr(dy), _ptr(dx), c_i(N), c_i(H), c_i(W_),
+                                c_i(C), _stream())
+    _check("downsample_a_bwd")
+    return dx
+
+
+def gap_fwd(x):
+    _bf16(x, "gap.x")
+    N, H, W_, C = x.shape
+    y = torch.empty(N, C, dtype=x.dtype, device=x.device)
+    _lib.cilfw_gap_fwd(_ptr(x), _ptr(y), c_i(N), c_i(H * W_), c_i(C),
+                       _stream())
+    _check("gap_fwd")
+    return y
+
+
+def gap_bwd(dy, H, W_):
+    N, C = dy.shape
+    dx = torch.empty(N, H, W_, C, dtype=dy.dtype, device=dy.device)
+    _lib.cilfw_gap_bwd(_ptr(dy), _ptr(dx), c_i(H * W_), c_i(C),
+                       c_l(N * H * W_ * C), _stream())
+    _check("gap_bwd")
+    return dx
+
+
+def maxpool_fwd(x, kernel, stride, pad):
+    _bf16(x, "maxpool.x")
+    N, H, W_, C = x.shape
+    Ho = (H + 2 * pad - kernel) // stride + 1
+    Wo = (W_ + 2 * pad - kernel) // stride + 1
+    y = torch.empty(N, Ho, Wo, C, dtype=x.dtype, device=x.device)
+    idx = torch.empty(N, Ho, Wo, C, dtype=torch.int32, device=x.device)
+    _lib.cilfw_maxpool_fwd(_ptr(x), _ptr(y), _ptr(idx), c_i(N), c_i(H),
+                           c_i(W_), c_i(C), c_i(kernel), c_i(stride),
+                           c_i(pad), c_i(Ho), c_i(Wo), _stream())
+    _check("maxpool_fwd")
+    return y, idx
+
+
+def maxpool_bwd(dy, idx, H, W_, kernel, stride, pad):
+    N, Ho, Wo, C = dy.shape
+    dx = torch.empty(N, H, W_, C, dtype=dy.dtype, device=dy.device)
+    _lib.cilfw_maxpool_bwd(_ptr(dy), _ptr(idx), _ptr(dx), c_i(N), c_i(H),
+                           c_i(W_), c_i(C), c_i(kernel), c_i(stride),
+                           c_i(pad), c_i(Ho), c_i(Wo), _stream())
+    _check("maxpool_bwd")
+    return dx
+
+
+# ---------------------------------------------------------------------- linear
+
+def linear_fwd(x, w, bias):
+    _bf16(x, "linear.x")
+    _bf16(w, "linear.w")
+    M, K = x.shape
+    N, _ = w.shape
+    y = torch.empty(M, N, dtype=torch.bfloat16, device=x.device)
+    bf = bias.float().contiguous() if bias is not None else None
+    _lib.cilfw_linear_fwd(_ptr(x), _ptr(w), _ptr(bf), _ptr(y), c_i(M), c_i(N),
+                          c_i(K), _stream())
+    _check("linear_fwd")
+    return y
+
+
+def linear_bwd(dy, x, w, has_bias):
+    _bf16(dy, "linear_bwd.dy")
+    M, K = x.shape
+    N, _ = w.shape
+    dx = torch.empty(M, K, dtype=torch.bfloat16, device=x.device)
+    dw = torch.empty(N, K, dtype=torch.float32, device=x.device)
+    db = torch.empty(N, dtype=torch.float32, device=x.device) if has_bias \
+        else None
+    _lib.cilfw_linear_dx(_ptr(dy), _ptr(w), _ptr(dx), c_i(M), c_i(N), c_i(K),
+                         _stream())
+    _lib.cilfw_linear_dw(_ptr(dy), _ptr(x), _ptr(dw), _ptr(db), c_i(M),
+                         c_i(N), c_i(K), _stream())
+    _check("linear_bwd")
+    return dx, dw, db
+
+
+# ---------------------------------------------------------------------- losses
+
+def ce_fwd(logits, targets, smooth):
+    logits = logits.float().contiguous()
+    M, C = logits.shape
+    probs = torch.empty_like(logits)
+    loss = torch.empty((), dtype=torch.float32, device=logits.device)
+    _lib.cilfw_ce_fwd(_ptr(logits), _ptr(targets.contiguous()), _ptr(probs),
+                      _ptr(loss), c_i(M), c_i(C), c_f(smooth), _stream())
+    _check("ce_fwd")
+    return loss, probs
+
+
+def ce_bwd(probs, targets, smooth, dloss):
+    M, C = probs.shape
+    dlogits = torch.empty_like(probs)
+    dl = dloss.float().reshape(1).contiguous()
+    _lib.cilfw_ce_bwd(_ptr(probs), _ptr(targets.contiguous()), _ptr(dl),
+                      _ptr(dlogits), c_i(M), c_i(C), c_f(smooth), _stream())
+    _check("ce_bwd")
+    return dlogits
+
+
+def kd_fwd(s_logits, t_logits, T):
+    s = s_logits.float().contiguous()
+    t = t_logits.float().contiguous()
+    M, C = s.shape
+    ps = torch.empty_like(s)
+    pt = torch.empty_like(s)
+    loss = torch.empty((), dtype=torch.float32, device=s.device)
+    _lib.cilfw_kd_fwd(_ptr(s), _ptr(t), _ptr(ps), _ptr(pt), _ptr(loss),
+                      c_i(M), c_i(C), c_f(T), _stream())
+    _check("kd_fwd")
+    return loss, ps, pt
+
+
+def kd_bwd(ps, pt, T, dloss):
+    M, C = ps.shape
+    ds = torch.empty_like(ps)
+    dl = dloss.float().reshape(1).contiguous()
+    _lib.cilfw_kd_bwd(_ptr(ps), _ptr(pt), _ptr(dl), _ptr(ds), c_i(M), c_i(C),
+                      c_f(T), _stream())
+    _check("kd_bwd")
+    return ds
+
+
+# ------------------------------------------------------------- optimizer / misc
+
+def sgd_step(p, g, m, lr, momentum, wd):
+    _lib.cilfw_sgd_step(_ptr(p), _ptr(g), _ptr(m), c_l(p.numel()), c_f(lr),
+                        c_f(momentum), c_f(wd), _stream())
+    _check("sgd_step")
+
+
+def topk_correct(logits, targets, maxk):
+    lf = logits.float().contiguous()
+    M, C = lf.shape
+    counts = torch.empty(maxk, dtype=torch.int64, device=lf.device)
+    _lib.cilfw_topk_correct(_ptr(lf), _ptr(targets.contiguous()),
+                            _ptr(counts), c_i(M), c_i(C), c_i(maxk),
+                            _stream())
+    _check("topk_correct")
+    return counts
+
+
+def herding_select(f, mu, m):
+    n, D = f.shape
+    assert (n + D) * 4 <= 60_000, \
+        "herding_select kernel supports n+D <= 15000 (per-class sample sets)"
+    order = torch.empty(m, dtype=torch.int64, device=f.device)
+    _lib.cilfw_herding_select(_ptr(f), _ptr(mu), _ptr(order), c_i(n), c_i(D),
+                              c_i(m), _stream())
+    _check("herding_select")
+    return order

@@ -1,0 +1,478 @@
+// cilfw — BatchNorm (fused ReLU), add+ReLU, DownsampleA, global-avg-pool and
+// max-pool kernels for gfx950. NHWC bf16 activations, fp32 statistics/params.
+// Replaces the reference's cuDNN BN + ATen elementwise kernels
+// (SURVEY.md §2.3 K3/K4/K5/K6).
+
+#include "common.h"
+
+#define NT 256
+
+// ---------------------------------------------------------------- BN statistics
+// pass 1: per-channel sum & sumsq via per-block partials + fp32 atomics.
+// x viewed as (M, C); grid.x covers C/64, grid.y covers row-chunks.
+
+__global__ __launch_bounds__(NT)
+void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ sum,
+                    float* __restrict__ sumsq, long M, int C, int rows_per_blk) {
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rlane = threadIdx.x >> 6;  // 0..3
+  const bool active = c < C;  // never early-return before __syncthreads
+  long r0 = (long)blockIdx.y * rows_per_blk + rlane;
+  long r1 = min((long)(blockIdx.y + 1) * rows_per_blk, M);
+  float s = 0.f, q = 0.f;
+  if (active)
+    for (long r = r0; r < r1; r += 4) {
+      float v = bf2f(x[r * C + c]);
+      s += v;
+      q += v * v;
+    }
+  // reduce the 4 row-lanes of this channel through LDS
+  __shared__ float red[2][4][64];
+  red[0][rlane][threadIdx.x & 63] = s;
+  red[1][rlane][threadIdx.x & 63] = q;
+  __syncthreads();
+  if (rlane == 0 && active) {
+    s = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
+        red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
+    q = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
+        red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
+    atomicAdd(&sum[c], s);
+    atomicAdd(&sumsq[c], q);
+  }
+}
+
+// pass 2: finalize mean/invstd (+ running stats update, training only)
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   long M, int C, float momentum, float eps,
+                                   int training) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  if (training) {
+    float mu = sum[c] / (float)M;
+    float var = fmaxf(sumsq[c] / (float)M - mu * mu, 0.f);
+    mean[c] = mu;
+    invstd[c] = rsqrtf(var + eps);
+    float unbiased = var * (float)M / (float)max(M - 1, 1L);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  } else {
+    mean[c] = running_mean[c];
+    invstd[c] = rsqrtf(running_var[c] + eps);
+  }
+}
+
+// pass 3: y = gamma*(x-mean)*invstd + beta (+relu). 8 bf16 per thread.
+__global__ __launch_bounds__(NT)
+void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+                     const float* __restrict__ gamma,
+                     const float* __restrict__ beta,
+                     const float* __restrict__ mean,
+                     const float* __restrict__ invstd,
+                     long total, int C, int relu) {
+  extern __shared__ float params[];  // [4][C]: gamma,beta,mean,invstd
+  for (int c = threadIdx.x; c < C; c += NT) {
+    params[c] = gamma[c];
+    params[C + c] = beta[c];
+    params[2 * C + c] = mean[c];
+    params[3 * C + c] = invstd[c];
+  }
+  __syncthreads();
+  long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  if (i0 + 8 > total) {
+    for (long i = i0; i < total; ++i) {
+      int c = (int)(i % C);
+      float v = (bf2f(x[i]) - params[2 * C + c]) * params[3 * C + c] *
+                params[c] + params[C + c];
+      if (relu) v = fmaxf(v, 0.f);
+      y[i] = f2bf(v);
+    }
+    return;
+  }
+  int4 xv = *(const int4*)&x[i0];
+  bf16_t* xe = (bf16_t*)&xv;
+  bf16_t out[8];
+  int c0 = (int)(i0 % C);  // C % 8 == 0 for all cilfw models
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c0 + j;
+    float v = (bf2f(xe[j]) - params[2 * C + c]) * params[3 * C + c] *
+              params[c] + params[C + c];
+    if (relu) v = fmaxf(v, 0.f);
+    out[j] = f2bf(v);
+  }
+  *(int4*)&y[i0] = *(int4*)out;
+}
+
+// backward pass 1: dbeta = sum dy', dgamma = sum dy'*xhat  (dy' relu-masked)
+__global__ __launch_bounds__(NT)
+void bn_bwd_sums_kernel(const bf16_t* __restrict__ dy,
+                        const bf16_t* __restrict__ x,
+                        const bf16_t* __restrict__ y,
+                        const float* __restrict__ mean,
+                        const float* __restrict__ invstd,
+                        float* __restrict__ dgamma, float* __restrict__ dbeta,
+                        long M, int C, int rows_per_blk, int relu) {
+  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+  const int rlane = threadIdx.x >> 6;
+  const bool active = c < C;
+  const float mu = active ? mean[c] : 0.f, is = active ? invstd[c] : 0.f;
+  long r0 = (long)blockIdx.y * rows_per_blk + rlane;
+  long r1 = min((long)(blockIdx.y + 1) * rows_per_blk, M);
+  float sg = 0.f, sb = 0.f;
+  if (active)
+    for (long r = r0; r < r1; r += 4) {
+      long i = r * C + c;
+      float g = bf2f(dy[i]);
+      if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
+      float xhat = (bf2f(x[i]) - mu) * is;
+      sb += g;
+      sg += g * xhat;
+    }
+  __shared__ float red[2][4][64];
+  red[0][rlane][threadIdx.x & 63] = sg;
+  red[1][rlane][threadIdx.x & 63] = sb;
+  __syncthreads();
+  if (rlane == 0 && active) {
+    sg = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
+         red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
+    sb = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
+         red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
+    atomicAdd(&dgamma[c], sg);
+    atomicAdd(&dbeta[c], sb);
+  }
+}
+
+// backward pass 2: dx = (gamma*invstd/M) * (M*dy' - dbeta - xhat*dgamma)
+__global__ __launch_bounds__(NT)
+void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
+                         const bf16_t* __restrict__ x,
+                         const bf16_t* __restrict__ y,
+                         bf16_t* __restrict__ dx,
+                         const float* __restrict__ gamma,
+                         const float* __restrict__ mean,
+                         const float* __restrict__ invstd,
+                         const float* __restrict__ dgamma,
+                         const float* __restrict__ dbeta,
+                         long total, long M, int C, int relu, int training) {
+  extern __shared__ float params[];  // [5][C]: gamma*invstd, mean, invstd, dg, db
+  for (int c = threadIdx.x; c < C; c += NT) {
+    params[c] = gamma[c] * invstd[c];
+    params[C + c] = mean[c];
+    params[2 * C + c] = invstd[c];
+    params[3 * C + c] = dgamma[c];
+    params[4 * C + c] = dbeta[c];
+  }
+  __syncthreads();
+  const float rM = 1.f / (float)M;
+  long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  long iend = min(i0 + 8, total);
+  for (long i = i0; i < iend; ++i) {
+    int c = (int)(i % C);
+    float g = bf2f(dy[i]);
+    if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
+    float v;
+    if (training) {
+      float xhat = (bf2f(x[i]) - params[C + c]) * params[2 * C + c];
+      v = params[c] * rM *
+          ((float)M * g - params[4 * C + c] - xhat * params[3 * C + c]);
+    } else {
+      v = g * params[c];
+    }
+    dx[i] = f2bf(v);
+  }
+}
+
+// ------------------------------------------------------------------- add+ReLU
+
+__global__ __launch_bounds__(NT)
+void add_relu_fwd_kernel(const bf16_t* __restrict__ a,
+                         const bf16_t* __restrict__ b,
+                         bf16_t* __restrict__ y, long total) {
+  long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  if (i0 >= total) return;
+  if (i0 + 8 <= total) {
+    int4 av = *(const int4*)&a[i0];
+    int4 bv = *(const int4*)&b[i0];
+    bf16_t* ae = (bf16_t*)&av;
+    bf16_t* be = (bf16_t*)&bv;
+    bf16_t out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out[j] = f2bf(fmaxf(bf2f(ae[j]) + bf2f(be[j]), 0.f));
+    *(int4*)&y[i0] = *(int4*)out;
+  } else {
+    for (long i = i0; i < total; ++i)
+      y[i] = f2bf(fmaxf(bf2f(a[i]) + bf2f(b[i]), 0.f));
+  }
+}
+
+__global__ __launch_bounds__(NT)
+void add_relu_bwd_kernel(const bf16_t* __restrict__ dy,
+                         const bf16_t* __restrict__ y,
+                         bf16_t* __restrict__ da, long total) {
+  long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
+  if (i0 >= total) return;
+  long iend = min(i0 + 8, total);
+  for (long i = i0; i < iend; ++i)
+    da[i] = (bf2f(y[i]) > 0.f) ? dy[i] : (bf16_t)0;
+}
+
+// ---------------------------------------------------------------- DownsampleA
+// y (N,H/2,W/2,2C): [:, :, :, :C] = x[:, ::2, ::2, :], rest zero.
+
+__global__ __launch_bounds__(NT)
+void downsample_a_fwd_kernel(const bf16_t* __restrict__ x,
+                             bf16_t* __restrict__ y, int N, int H, int W,
+                             int C, long total_out) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_out) return;
+  int C2 = 2 * C, Ho = H / 2, Wo = W / 2;
+  int c = (int)(i % C2);
+  long rest = i / C2;
+  int wo = (int)(rest % Wo);
+  rest /= Wo;
+  int ho = (int)(rest % Ho);
+  int n = (int)(rest / Ho);
+  bf16_t v = 0;
+  if (c < C)
+    v = x[(((long)n * H + 2 * ho) * W + 2 * wo) * C + c];
+  y[i] = v;
+}
+
+__global__ __launch_bounds__(NT)
+void downsample_a_bwd_kernel(const bf16_t* __restrict__ dy,
+                             bf16_t* __restrict__ dx, int N, int H, int W,
+                             int C, long total_in) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_in) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int w = (int)(rest % W);
+  rest /= W;
+  int h = (int)(rest % H);
+  int n = (int)(rest / H);
+  bf16_t v = 0;
+  if ((h & 1) == 0 && (w & 1) == 0) {
+    int Ho = H / 2, Wo = W / 2;
+    v = dy[(((long)n * Ho + h / 2) * Wo + w / 2) * (2 * C) + c];
+  }
+  dx[i] = v;
+}
+
+// ------------------------------------------------------------ global avg pool
+
+__global__ __launch_bounds__(NT)
+void gap_fwd_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+                    int HW, int C) {
+  const int n = blockIdx.x;
+  const float scale = 1.f / (float)HW;
+  for (int c = threadIdx.x; c < C; c += NT) {
+    float s = 0.f;
+    const bf16_t* base = x + (long)n * HW * C + c;
+    for (int i = 0; i < HW; ++i) s += bf2f(base[(long)i * C]);
+    y[(long)n * C + c] = f2bf(s * scale);
+  }
+}
+
+__global__ __launch_bounds__(NT)
+void gap_bwd_kernel(const bf16_t* __restrict__ dy, bf16_t* __restrict__ dx,
+                    int HW, int C, long total) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total) return;
+  int c = (int)(i % C);
+  long n = i / ((long)HW * C);
+  dx[i] = f2bf(bf2f(dy[n * C + c]) / (float)HW);
+}
+
+// ------------------------------------------------------------------- max pool
+
+__global__ __launch_bounds__(NT)
+void maxpool_fwd_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+                        int* __restrict__ idx, int N, int H, int W, int C,
+                        int kk, int st, int pad, int Ho, int Wo,
+                        long total_out) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_out) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int wo = (int)(rest % Wo);
+  rest /= Wo;
+  int ho = (int)(rest % Ho);
+  int n = (int)(rest / Ho);
+  float best = -3.4e38f;
+  int bi = 0;
+  for (int r = 0; r < kk; ++r) {
+    int hi = ho * st - pad + r;
+    if (hi < 0 || hi >= H) continue;
+    for (int s = 0; s < kk; ++s) {
+      int wi = wo * st - pad + s;
+      if (wi < 0 || wi >= W) continue;
+      float v = bf2f(x[(((long)n * H + hi) * W + wi) * C + c]);
+      if (v > best) { best = v; bi = hi * W + wi; }
+    }
+  }
+  y[i] = f2bf(best);
+  idx[i] = bi;
+}
+
+__global__ __launch_bounds__(NT)
+void maxpool_bwd_kernel(const bf16_t* __restrict__ dy,
+                        const int* __restrict__ idx, bf16_t* __restrict__ dx,
+                        int N, int H, int W, int C, int kk, int st, int pad,
+                        int Ho, int Wo, long total_in) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_in) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int wi = (int)(rest % W);
+  rest /= W;
+  int hi = (int)(rest % H);
+  int n = (int)(rest / H);
+  int my = hi * W + wi;
+  float acc = 0.f;
+  // windows whose output could have selected (hi, wi)
+  int ho_lo = max(0, (hi + pad - kk + st) / st), ho_hi = min(Ho - 1,
+                                                            (hi + pad) / st);
+  int wo_lo = max(0, (wi + pad - kk + st) / st), wo_hi = min(Wo - 1,
+                                                            (wi + pad) / st);
+  for (int ho = ho_lo; ho <= ho_hi; ++ho)
+    for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+      long o = (((long)n * Ho + ho) * Wo + wo) * C + c;
+      if (idx[o] == my) acc += bf2f(dy[o]);
+    }
+  dx[i] = f2bf(acc);
+}
+
+// ============================== launchers ==============================
+
+extern "C" {
+
+void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
+                  void* running_mean, void* running_var, void* mean,
+                  void* invstd, void* scratch_sums, long M, int C,
+                  float momentum, float eps, int training, int relu,
+                  void* stream) {
+  hipStream_t st = (hipStream_t)stream;
+  if (training) {
+    hipMemsetAsync(scratch_sums, 0, 2 * C * sizeof(float), st);
+    float* sum = (float*)scratch_sums;
+    float* sumsq = sum + C;
+    int rows_per_blk = (int)min((long)4096, max((long)256, M / 64));
+    dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
+    hipLaunchKernelGGL(bn_sums_kernel, grid, dim3(NT), 0, st,
+                       (const bf16_t*)x, sum, sumsq, M, C, rows_per_blk);
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
+                       st, sum, sumsq, (float*)mean, (float*)invstd,
+                       (float*)running_mean, (float*)running_var, M, C,
+                       momentum, eps, 1);
+  } else {
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
+                       st, nullptr, nullptr, (float*)mean, (float*)invstd,
+                       (float*)running_mean, (float*)running_var, M, C,
+                       momentum, eps, 0);
+  }
+  long total = M * C;
+  long blocks = cdiv((long)total, (long)NT * 8);
+  hipLaunchKernelGGL(bn_apply_kernel, dim3((int)blocks), dim3(NT),
+                     4 * C * sizeof(float), st, (const bf16_t*)x, (bf16_t*)y,
+                     (const float*)gamma, (const float*)beta,
+                     (const float*)mean, (const float*)invstd, total, C, relu);
+}
+
+void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
+                  const void* gamma, const void* mean, const void* invstd,
+                  void* dgamma, void* dbeta, long M, int C, int relu,
+                  int training, void* stream) {
+  hipStream_t st = (hipStream_t)stream;
+  hipMemsetAsync(dgamma, 0, C * sizeof(float), st);
+  hipMemsetAsync(dbeta, 0, C * sizeof(float), st);
+  int rows_per_blk = (int)min((long)4096, max((long)256, M / 64));
+  dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
+  hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
+                     (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
+                     (const float*)mean, (const float*)invstd, (float*)dgamma,
+                     (float*)dbeta, M, C, rows_per_blk, relu);
+  long total = M * C;
+  long blocks = cdiv((long)total, (long)NT * 8);
+  hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),
+                     5 * C * sizeof(float), st, (const bf16_t*)dy,
+                     (const bf16_t*)x, (const bf16_t*)y, (bf16_t*)dx,
+                     (const float*)gamma, (const float*)mean,
+                     (const float*)invstd, (const float*)dgamma,
+                     (const float*)dbeta, total, M, C, relu, training);
+}
+
+void cilfw_add_relu_fwd(const void* a, const void* b, void* y, long total,
+                        void* stream) {
+  long blocks = cdiv((long)total, (long)NT * 8);
+  hipLaunchKernelGGL(add_relu_fwd_kernel, dim3((int)blocks), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)a, (const bf16_t*)b,
+                     (bf16_t*)y, total);
+}
+
+void cilfw_add_relu_bwd(const void* dy, const void* y, void* da, long total,
+                        void* stream) {
+  long blocks = cdiv((long)total, (long)NT * 8);
+  hipLaunchKernelGGL(add_relu_bwd_kernel, dim3((int)blocks), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)y,
+                     (bf16_t*)da, total);
+}
+
+void cilfw_downsample_a_fwd(const void* x, void* y, int N, int H, int W,
+                            int C, void* stream) {
+  long total = (long)N * (H / 2) * (W / 2) * 2 * C;
+  hipLaunchKernelGGL(downsample_a_fwd_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y, N, H,
+                     W, C, total);
+}
+
+void cilfw_downsample_a_bwd(const void* dy, void* dx, int N, int H, int W,
+                            int C, void* stream) {
+  long total = (long)N * H * W * C;
+  hipLaunchKernelGGL(downsample_a_bwd_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy, (bf16_t*)dx, N,
+                     H, W, C, total);
+}
+
+void cilfw_gap_fwd(const void* x, void* y, int N, int HW, int C,
+                   void* stream) {
+  hipLaunchKernelGGL(gap_fwd_kernel, dim3(N), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y, HW, C);
+}
+
+void cilfw_gap_bwd(const void* dy, void* dx, int N, int HW, int C,
+                   void* stream) {
+  long total = (long)N * HW * C;
+  hipLaunchKernelGGL(gap_bwd_kernel, dim3((int)cdiv((long)total, (long)NT)),
+                     dim3(NT), 0, (hipStream_t)stream, (const bf16_t*)dy,
+                     (bf16_t*)dx, HW, C, total);
+}
+
+void cilfw_maxpool_fwd(const void* x, void* y, void* idx, int N, int H, int W,
+                       int C, int kk, int st, int pad, int Ho, int Wo,
+                       void* stream) {
+  long total = (long)N * Ho * Wo * C;
+  hipLaunchKernelGGL(maxpool_fwd_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)y,
+                     (int*)idx, N, H, W, C, kk, st, pad, Ho, Wo, total);
+}
+
+void cilfw_maxpool_bwd(const void* dy, const void* idx, void* dx, int N,
+                       int H, int W, int C, int kk, int st, int pad, int Ho,
+                       int Wo, void* stream) {
+  long total = (long)N * H * W * C;
+  hipLaunchKernelGGL(maxpool_bwd_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)dy, (const int*)idx,
+                     (bf16_t*)dx, N, H, W, C, kk, st, pad, Ho, Wo, total);
+}
+
+}  // extern "C"

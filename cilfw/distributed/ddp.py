@@ -1,0 +1,144 @@
+"""First-party data-parallel engine: flat params + bucketed, backward-overlapped
+gradient all-reduce on RCCL over xGMI.
+
+Replaces torch DDP (reference template.py:243-244). Design:
+- ALL trainable params are coalesced into ONE flat fp32 buffer (param order
+  reversed so the flat layout matches gradient-readiness order during backward);
+  grads live in a matching flat buffer. The fused SGD step then runs as a single
+  kernel over the flat buffers, and a full-model broadcast is one collective.
+- The flat grad buffer is cut into >= bucket_mb-sized buckets. A
+  post-accumulate-grad hook counts arrivals per bucket; when a bucket's params have
+  all produced grads, an async all-reduce is launched on that slice — RCCL runs it
+  on its own stream, overlapping communication with the rest of backward. xGMI
+  links are point-to-point (7 x ~153 GB/s per GPU), so for the small messages of
+  this workload (2-100 MB of grads) fewer, larger buckets are preferred — default
+  25 MB means 1-4 buckets for the models here.
+- finalize() flushes remaining buckets, waits on all works, and averages.
+
+Works with gloo on CPU (tests) and RCCL on ROCm; with world_size == 1 the engine
+still provides flattening (no collectives).
+"""
+
+import torch
+import torch.distributed as dist
+
+
+def _dist_active():
+    return dist.is_available() and dist.is_initialized() and \
+        dist.get_world_size() > 1
+
+
+class DataParallelEngine:
+    def __init__(self, model, bucket_mb=25.0, process_group=None):
+        self.model = model
+        self.group = process_group
+        self.params = [p for p in model.parameters() if p.requires_grad]
+        # reverse: classifier/head params (constructed last) produce grads first
+        self.params = self.params[::-1]
+        assert all(p.dtype == torch.float32 for p in self.params), \
+            "cilfw keeps fp32 master params; compute casts are per-op"
+        dev = self.params[0].device
+        self.numel = sum(p.numel() for p in self.params)
+        self.flat_params = torch.empty(self.numel, dtype=torch.float32, device=dev)
+        self.flat_grads = torch.zeros(self.numel, dtype=torch.float32, device=dev)
+
+        # move params into the flat buffer; attach grad views
+        offset = 0
+        self._offsets = []
+        for p in self.params:
+            n = p.numel()
+            self.flat_params[offset:offset + n].copy_(p.data.reshape(-1))
+            p.data = self.flat_params[offset:offset + n].view_as(p)
+            p.grad = self.flat_grads[offset:offset + n].view_as(p)
+            self._offsets.append(offset)
+            offset += n
+
+        # bucket boundaries (by element count)
+        bucket_elems = max(int(bucket_mb * 1024 * 1024 / 4), 1)
+        self.buckets = []          # list of (start, end, param_indices)
+        start, count, idxs = 0, 0, []
+        for i, p in enumerate(self.params):
+            idxs.append(i)
+            count += p.numel()
+            if count >= bucket_elems:
+                self.buckets.append((start, start + count, tuple(idxs)))
+                start += count
+                count, idxs = 0, []
+        if idxs:
+            self.buckets.append((start, start + count, tuple(idxs)))
+        self._param_bucket = {}
+        for bi, (_, _, idxs) in enumerate(self.buckets):
+            for i in idxs:
+                self._param_bucket[i] = bi
+
+        self._arrived = [0] * len(self.buckets)
+        self._launched = [False] * len(self.buckets)
+        self._works = []
+        self._hooks = []
+        self._hook_enabled = True
+        for i, p in enumerate(self.params):
+            self._hooks.append(p.register_post_accumulate_grad_hook(
+                self._make_hook(i)))
+
+        # replicate rank-0 initial weights (reference: DDP broadcast at wrap,
+        # SURVEY.md §2.3 N6) — one collective for the whole model
+        if _dist_active():
+            dist.broadcast(self.flat_params, src=0, group=self.group)
+
+    def _make_hook(self, i):
+        bi = self._param_bucket[i]
+
+        def hook(_p):
+            if not self._hook_enabled:
+                return
+            self._arrived[bi] += 1
+            if self._arrived[bi] == len(self.buckets[bi][2]):
+                self._launch(bi)
+        return hook
+
+    def _launch(self, bi):
+        if self._launched[bi] or not _dist_active():
+            self._launched[bi] = True
+            return
+        s, e, _ = self.buckets[bi]
+        work = dist.all_reduce(self.flat_grads[s:e], op=dist.ReduceOp.SUM,
+                               group=self.group, async_op=True)
+        self._works.append(work)
+        self._launched[bi] = True
+
+    def zero_grad(self):
+        self.flat_grads.zero_()
+        self._arrived = [0] * len(self.buckets)
+        self._launched = [False] * len(self.buckets)
+        self._works = []
+
+    def finalize(self):
+        """Call after backward: flush stragglers, wait, average."""
+        for bi in range(len(self.buckets)):
+            if not self._launched[bi]:
+                self._launch(bi)
+        for w in self._works:
+            w.wait()
+        self._works = []
+        if _dist_active():
+            self.flat_grads.mul_(1.0 / dist.get_world_size(group=self.group))
+
+    def no_sync(self):
+        """Context manager disabling the hooks (grad accumulation)."""
+        engine = self
+
+        class _NoSync:
+            def __enter__(self):
+                engine._hook_enabled = False
+
+            def __exit__(self, *a):
+                engine._hook_enabled = True
+        return _NoSync()
+
+    def detach(self):
+        """Remove hooks and unflatten (used before re-wrapping for a new task)."""
+        for h in self._hooks:
+            h.remove()
+        for p in self.params:
+            p.data = p.data.clone()
+            p.grad = None

@@ -1,0 +1,224 @@
+"""CIL orchestrator — the per-task lifecycle (reference template.py:191-303).
+
+Task loop: grow classifier -> wrap in the cilfw DP engine -> train (CE + KD vs the
+frozen previous-task teacher) -> weight-align the new head -> cumulative eval ->
+snapshot teacher -> herding feature pass -> update rehearsal memory -> checkpoint.
+
+Printed schema matches the reference so runs are comparable:
+per-epoch train meters, "* Acc@1 ... loss ..." eval lines, per-task
+"task id = ...  @Acc1 = ..., acc1s = [...]".
+"""
+
+import random
+import time
+
+import numpy as np
+import torch
+from torch.utils.data import DataLoader
+
+from . import ops
+from .data import build_dataset, DistributedSampler
+from .data.transforms import EvalTransform
+from .distributed import (init_distributed_mode, DataParallelEngine, barrier,
+                          get_world_size, get_rank)
+from .models import CilModel
+from .cil import RehearsalMemory, save_task_checkpoint, load_task_checkpoint
+from .optim import FlatSGD, CosineLR
+from .utils.metrics import MetricLogger, SmoothedValue
+
+
+def init_seed(args):
+    seed = args.seed
+    random.seed(seed)
+    np.random.seed(seed)
+    torch.manual_seed(seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(seed)
+
+
+def compute_dtype(args):
+    if args.dtype == "bf16" and str(args.device).startswith("cuda"):
+        return torch.bfloat16
+    return torch.float32
+
+
+def _to_device(inputs, targets, device, dtype):
+    inputs = inputs.to(device, non_blocking=True).to(dtype)
+    targets = targets.to(device, non_blocking=True)
+    return inputs, targets
+
+
+@torch.no_grad()
+def evaluate(model, loader, device, args, header="Test:"):
+    model.eval()
+    dtype = compute_dtype(args)
+    metric_logger = MetricLogger()
+    nb_classes = model.fc.nb_classes
+    topk = (1, min(5, nb_classes))
+    for inputs, targets, _tids in loader:
+        inputs, targets = _to_device(inputs, targets, device, dtype)
+        logits, _ = model(inputs)
+        loss = ops.cross_entropy(logits.float(), targets)
+        accs = ops.accuracy(logits, targets, topk=topk)
+        bs = targets.shape[0]
+        metric_logger.update(loss=loss.item())
+        metric_logger.update_n(n=bs, acc1=accs[0], acc5=accs[1])
+    metric_logger.synchronize_between_processes(device=torch.device(device))
+    acc1 = metric_logger.meters["acc1"].global_avg
+    acc5 = metric_logger.meters["acc5"].global_avg
+    lossv = metric_logger.meters["loss"].global_avg
+    print(f"* Acc@1 {acc1:.3f} Acc@5 {acc5:.3f} loss {lossv:.3f}")
+    model.train()
+    return acc1
+
+
+def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
+                   train_sampler, val_loader, device, args):
+    dtype = compute_dtype(args)
+    known = args.known_classes
+    lambda_kd = args.lambda_kd
+    if args.dynamic_lambda_kd and known > 0:
+        # n/(n+m) scaling the reference documented but never wired
+        # (README.md:175-176 / dead flag template.py:48)
+        lambda_kd = known / (known + args.increment_per_task)
+    model.train()
+    for epoch in range(args.num_epochs):
+        if train_sampler is not None:
+            train_sampler.set_epoch(epoch)
+        metric_logger = MetricLogger()
+        metric_logger.meters["lr"] = SmoothedValue(fmt="{value:.6f}")
+        t0 = time.time()
+        nimg = 0
+        for inputs, targets, _tids in train_loader:
+            inputs, targets = _to_device(inputs, targets, device, dtype)
+            optimizer.zero_grad()
+            logits, _features = model(inputs)
+            loss_ce = ops.cross_entropy(logits.float(), targets, args.smooth)
+            if teacher is not None:
+                with torch.no_grad():
+                    t_logits, _ = teacher(inputs)
+                loss_kd = ops.kd_loss(logits[:, :known].float(),
+                                      t_logits.float(), args.kd_temperature)
+                loss = loss_ce + lambda_kd * loss_kd
+            else:
+                loss_kd = torch.zeros((), device=logits.device)
+                loss = loss_ce
+            loss.backward()
+            engine.finalize()
+            optimizer.step()
+            if args.compat_step_barrier:
+                barrier()  # reference's per-step barrier (template.py:272)
+            accs = ops.accuracy(logits, targets,
+                                topk=(1, min(5, logits.shape[1])))
+            bs = targets.shape[0]
+            nimg += bs
+            metric_logger.update(ce=loss_ce.item(), kd=loss_kd.item(),
+                                 loss=loss.item())
+            metric_logger.update_n(n=bs, acc1=accs[0])
+            metric_logger.meters["lr"].update(optimizer.lr)
+        metric_logger.synchronize_between_processes(device=torch.device(device))
+        scheduler.step()
+        ips = nimg * get_world_size() / max(time.time() - t0, 1e-9)
+        print(f"task {args.task_id} epoch {epoch}: {metric_logger}  "
+              f"imgs/s {ips:.0f}")
+        if args.eval_every_epoch and (epoch + 1) % int(args.eval_every_epoch) == 0 \
+                and epoch + 1 < args.num_epochs:
+            evaluate(model, val_loader, device, args)
+
+
+@torch.no_grad()
+def extract_task_features(model, dataset, device, args):
+    """Unshuffled, NON-distributed feature pass over the full task set — the
+    reference runs this replicated on every rank (template.py:292-299); herding is
+    deterministic so the resulting memory is rank-identical."""
+    model.eval()
+    dtype = compute_dtype(args)
+    loader = DataLoader(dataset, batch_size=args.batch_size, shuffle=False,
+                        num_workers=args.workers, drop_last=False)
+    feats = []
+    for inputs, _targets, _tids in loader:
+        inputs = inputs.to(device, non_blocking=True).to(dtype)
+        feats.append(model.extract_vector(inputs).float())
+    model.train()
+    return torch.cat(feats)
+
+
+def run(args):
+    init_distributed_mode(args)
+    init_seed(args)
+    device = args.device
+
+    scenario_train, nb_classes = build_dataset(is_train=True, args=args)
+    scenario_val, _ = build_dataset(is_train=False, args=args)
+    args.nb_classes = nb_classes
+
+    model = CilModel(args.backbone, args.input_size).to(device)
+    barrier()
+
+    memory = RehearsalMemory(args.memory_size, args.herding_method,
+                             args.fixed_memory, nb_total_classes=nb_classes)
+    teacher = None
+    acc1s = []
+    args.known_classes = 0
+    start_task = 0
+
+    if args.resume:
+        state = load_task_checkpoint(args.resume, model, memory, args)
+        model = model.to(device)
+        acc1s = state["acc1s"]
+        start_task = state["task_id"] + 1
+        if start_task < len(scenario_train):
+            teacher = model.copy().to(device)
+            teacher.freeze(["all"])
+        print(f"resumed from {args.resume}: start_task={start_task}, "
+              f"known={args.known_classes}, acc1s={acc1s}")
+
+    for task_id in range(start_task, len(scenario_train)):
+        args.task_id = task_id
+        dataset_train = scenario_train[task_id]
+        dataset_val = scenario_val[:task_id + 1]
+        args.increment_per_task = scenario_train.increments(task_id)
+
+        if task_id > 0:
+            mx, my, mt = memory.get()
+            dataset_train.add_samples(mx, my, mt)
+
+        world, rank = get_world_size(), get_rank()
+        train_sampler = DistributedSampler(dataset_train, world, rank,
+                                           shuffle=True, seed=args.seed)
+        val_sampler = DistributedSampler(dataset_val, world, rank, shuffle=False)
+        train_loader = DataLoader(dataset_train, batch_size=args.batch_size,
+                                  sampler=train_sampler, num_workers=args.workers,
+                                  drop_last=True, persistent_workers=args.workers > 0)
+        val_loader = DataLoader(dataset_val, batch_size=args.batch_size,
+                                sampler=val_sampler, num_workers=args.workers)
+
+        model.prev_model_adaption(args.increment_per_task)
+        engine = DataParallelEngine(model, bucket_mb=args.ddp_bucket_mb)
+        optimizer = FlatSGD(engine, args.lr, args.momentum, args.weight_decay)
+        scheduler = CosineLR(optimizer, t_max=args.num_epochs)
+
+        train_one_task(model, teacher, engine, optimizer, scheduler,
+                       train_loader, train_sampler, val_loader, device, args)
+
+        model.after_model_adaption(args.increment_per_task, args)
+        acc1 = evaluate(model, val_loader, device, args)
+        acc1s.append(acc1)
+        print(f"task id = {task_id}  @Acc1 = {acc1:.5f}, acc1s = {acc1s}")
+
+        teacher = model.copy()
+        teacher.freeze(["all"])
+
+        features = extract_task_features(model, dataset_train, device, args)
+        rx, ry, rt = dataset_train.get_raw_samples()
+        memory.add(rx, ry, rt, features)
+
+        engine.detach()
+        args.known_classes += args.increment_per_task  # before snapshot: the
+        # checkpoint records the post-task state so resume starts task t+1
+        save_task_checkpoint(args.output_dir, task_id, model, memory, acc1s,
+                             args, optimizer, scheduler)
+
+    avg_inc_acc = sum(acc1s) / len(acc1s) if acc1s else 0.0
+    print(f"average incremental accuracy = {avg_inc_acc:.5f}")
+    return acc1s

@@ -1,0 +1,425 @@
+"""cilfw functional ops: autograd Functions with device dispatch.
+
+Conventions (cilfw-native, chosen for CDNA4 — not the reference's):
+- Activations are NHWC contiguous: ``(N, H, W, C)``. Channels innermost gives
+  coalesced per-channel access for BN stats and a dense GEMM K-dim (C*R*S) for the
+  implicit-GEMM MFMA convolutions.
+- Conv weights are ``(R, S, C, K)`` fp32 masters; forward casts to the compute dtype
+  (bf16) once per use. Weight grads come back fp32.
+- CPU tensors run the torch reference implementations (fp32 math) — these double as
+  the numerics oracles for the HIP kernels (tests/test_ops_gpu.py).
+- CUDA tensors run the hand-written HIP kernels via ``cilfw._hip_ops`` and raise if
+  the extension is missing (no silent ATen fallback).
+
+Reference behavior being reimplemented: the implicit cuDNN /
+cuBLAS kernels behind reference ``resnet.py`` and ``template.py:99-101, 259-263``.
+"""
+
+import torch
+import torch.nn.functional as F
+
+from ._backend import use_hip, ext
+
+
+def _to_nchw(x):
+    return x.permute(0, 3, 1, 2).contiguous()
+
+
+def _to_nhwc(x):
+    return x.permute(0, 2, 3, 1).contiguous()
+
+
+# --------------------------------------------------------------------------- conv2d
+
+
+class Conv2dNHWC(torch.autograd.Function):
+    """2D convolution, NHWC activations, (R,S,C,K) weight, symmetric padding."""
+
+    @staticmethod
+    def forward(ctx, x, w, stride, padding):
+        ctx.stride, ctx.padding = stride, padding
+        wc = w.to(x.dtype)
+        ctx.save_for_backward(x, wc)
+        ctx.w_dtype = w.dtype
+        if use_hip(x):
+            return ext().conv2d_fwd(x, wc, stride, padding)
+        # CPU reference: fp32 NCHW conv
+        xf = _to_nchw(x).float()
+        wf = w.permute(3, 2, 0, 1).contiguous().float()  # (K,C,R,S)
+        y = F.conv2d(xf, wf, stride=stride, padding=padding)
+        return _to_nhwc(y).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wc = ctx.saved_tensors
+        stride, padding = ctx.stride, ctx.padding
+        dy = dy.contiguous()
+        if use_hip(dy):
+            dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1], x.shape[2])
+            dw = ext().conv2d_bwd_weight(dy, x, stride, padding,
+                                         wc.shape[0], wc.shape[1])
+            return dx, dw.to(ctx.w_dtype), None, None
+        xf = _to_nchw(x).float()
+        wf = wc.permute(3, 2, 0, 1).contiguous().float()
+        dyf = _to_nchw(dy).float()
+        dxf = torch.nn.grad.conv2d_input(xf.shape, wf, dyf, stride=stride,
+                                         padding=padding)
+        dwf = torch.nn.grad.conv2d_weight(xf, wf.shape, dyf, stride=stride,
+                                          padding=padding)
+        dx = _to_nhwc(dxf).to(x.dtype)
+        dw = dwf.permute(2, 3, 1, 0).contiguous().to(ctx.w_dtype)  # (R,S,C,K)
+        return dx, dw, None, None
+
+
+def conv2d(x, w, stride=1, padding=1):
+    return Conv2dNHWC.apply(x, w, stride, padding)
+
+
+# ----------------------------------------------------------------- batchnorm (+ReLU)
+
+
+class BatchNormAct(torch.autograd.Function):
+    """Training-mode BN over NHWC (per-channel stats on N*H*W), optional fused ReLU.
+
+    Matches torch BN semantics: biased variance for normalization, unbiased for the
+    running-stat update; per-rank statistics (the reference runs plain BN under DDP —
+    no SyncBN).
+    """
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
+                training, relu):
+        ctx.relu = relu
+        ctx.eps = eps
+        if use_hip(x):
+            y, save_mean, save_invstd = ext().bn_fwd(
+                x, gamma, beta, running_mean, running_var,
+                momentum, eps, training, relu)
+            ctx.training = training
+            ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+            return y
+        N, H, W, C = x.shape
+        xf = x.float().reshape(-1, C)
+        if training:
+            mean = xf.mean(dim=0)
+            var = xf.var(dim=0, unbiased=False)
+            m = xf.shape[0]
+            with torch.no_grad():
+                running_mean.mul_(1 - momentum).add_(momentum * mean)
+                unbiased = var * (m / max(m - 1, 1))
+                running_var.mul_(1 - momentum).add_(momentum * unbiased)
+        else:
+            mean, var = running_mean.float(), running_var.float()
+        invstd = torch.rsqrt(var + eps)
+        yf = (xf - mean) * invstd * gamma.float() + beta.float()
+        if relu:
+            yf = yf.clamp_min(0)
+        y = yf.reshape(N, H, W, C).to(x.dtype)
+        ctx.training = training
+        ctx.save_for_backward(x, gamma, mean, invstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        relu = ctx.relu
+        dy = dy.contiguous()
+        if use_hip(dy):
+            dx, dgamma, dbeta = ext().bn_bwd(dy, x, gamma, mean, invstd, y,
+                                             relu, ctx.training)
+            return (dx, dgamma, dbeta) + (None,) * 6
+        C = x.shape[-1]
+        dyf = dy.float().reshape(-1, C)
+        if relu:
+            dyf = dyf * (y.float().reshape(-1, C) > 0)
+        xf = x.float().reshape(-1, C)
+        xhat = (xf - mean) * invstd
+        dgamma = (dyf * xhat).sum(dim=0)
+        dbeta = dyf.sum(dim=0)
+        if ctx.training:
+            m = xf.shape[0]
+            dxf = (gamma.float() * invstd / m) * (
+                m * dyf - dbeta - xhat * dgamma)
+        else:
+            dxf = dyf * gamma.float() * invstd
+        dx = dxf.reshape_as(x).to(x.dtype)
+        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype)) + (None,) * 6
+
+
+def batchnorm_act(x, gamma, beta, running_mean, running_var, momentum=0.1, eps=1e-5,
+                  training=True, relu=False):
+    return BatchNormAct.apply(x, gamma, beta, running_mean, running_var, momentum,
+                              eps, training, relu)
+
+
+# ------------------------------------------------------------------------- add+ReLU
+
+
+class AddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        if use_hip(a):
+            y = ext().add_relu_fwd(a, b)
+        else:
+            y = (a.float() + b.float()).clamp_min(0).to(a.dtype)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_hip(dy):
+            da = ext().add_relu_bwd(dy, y)
+        else:
+            da = (dy * (y > 0)).to(dy.dtype)
+        return da, da
+
+
+def add_relu(a, b):
+    return AddReLU.apply(a, b)
+
+
+# ------------------------------------------------------- downsample-A (reference C9)
+
+
+class DownsampleA(torch.autograd.Function):
+    """Stride-2 1x1 avg-pool + zero-channel pad (reference resnet.py:9-17):
+    (N,H,W,C) -> (N,H/2,W/2,2C) where the second C channels are zeros."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_shape = x.shape
+        if use_hip(x):
+            return ext().downsample_a_fwd(x)
+        N, H, W, C = x.shape
+        y = x.new_zeros(N, H // 2, W // 2, 2 * C)
+        y[:, :, :, :C] = x[:, ::2, ::2, :]
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        N, H, W, C = ctx.in_shape
+        dy = dy.contiguous()
+        if use_hip(dy):
+            return ext().downsample_a_bwd(dy, H, W)
+        dx = dy.new_zeros(N, H, W, C)
+        dx[:, ::2, ::2, :] = dy[:, :, :, :C]
+        return dx
+
+
+def downsample_a(x):
+    return DownsampleA.apply(x)
+
+
+# --------------------------------------------------------------- global average pool
+
+
+class GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_shape = x.shape
+        if use_hip(x):
+            return ext().gap_fwd(x)
+        N, H, W, C = x.shape
+        return x.float().mean(dim=(1, 2)).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        N, H, W, C = ctx.in_shape
+        dy = dy.contiguous()
+        if use_hip(dy):
+            return ext().gap_bwd(dy, H, W)
+        scale = 1.0 / (H * W)
+        return (dy * scale).reshape(N, 1, 1, C).expand(N, H, W, C).to(dy.dtype)
+
+
+def global_avg_pool(x):
+    return GlobalAvgPool.apply(x)
+
+
+# --------------------------------------------------------------------------- linear
+
+
+class LinearFn(torch.autograd.Function):
+    """y = x @ w^T + b with fp32 master w/b; grads fp32."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        wc = w.to(x.dtype)
+        ctx.save_for_backward(x, wc)
+        ctx.w_dtype = w.dtype
+        ctx.has_bias = b is not None
+        if use_hip(x):
+            return ext().linear_fwd(x, wc, b.float() if b is not None else None)
+        y = x.float() @ w.float().t()
+        if b is not None:
+            y = y + b.float()
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wc = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_hip(dy):
+            dx, dw, db = ext().linear_bwd(dy, x, wc, ctx.has_bias)
+            return dx, dw.to(ctx.w_dtype), (db.to(ctx.w_dtype) if ctx.has_bias
+                                            else None)
+        dyf = dy.float()
+        dx = (dyf @ wc.float()).to(x.dtype)
+        dw = (dyf.t() @ x.float()).to(ctx.w_dtype)
+        db = dyf.sum(dim=0).to(ctx.w_dtype) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def linear(x, w, b=None):
+    return LinearFn.apply(x, w, b)
+
+
+# ---------------------------------------------------- fused cross-entropy (+ smooth)
+
+
+class CrossEntropyLS(torch.autograd.Function):
+    """Fused log-softmax + NLL with label smoothing; mean over batch.
+
+    Reference: torch.nn.CrossEntropyLoss(label_smoothing=args.smooth) at
+    template.py:219,259.
+    """
+
+    @staticmethod
+    def forward(ctx, logits, targets, smoothing):
+        ctx.smoothing = smoothing
+        if use_hip(logits):
+            loss, probs = ext().ce_fwd(logits, targets, smoothing)
+            ctx.save_for_backward(probs, targets)
+            return loss
+        lf = logits.float()
+        logp = F.log_softmax(lf, dim=1)
+        n, c = lf.shape
+        nll = -logp.gather(1, targets.view(-1, 1)).squeeze(1)
+        if smoothing > 0:
+            loss = ((1 - smoothing) * nll - smoothing * logp.mean(dim=1)).mean()
+        else:
+            loss = nll.mean()
+        ctx.save_for_backward(logp.exp(), targets)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        probs, targets = ctx.saved_tensors
+        s = ctx.smoothing
+        if use_hip(probs):
+            dlogits = ext().ce_bwd(probs, targets, s, dloss)
+            return dlogits, None, None
+        n, c = probs.shape
+        g = probs.clone()
+        g.scatter_add_(1, targets.view(-1, 1), torch.full_like(targets, -1,
+                       dtype=g.dtype).view(-1, 1) * (1 - s))
+        if s > 0:
+            g -= s / c
+        dlogits = (g * (dloss / n)).to(probs.dtype)
+        return dlogits, None, None
+
+
+def cross_entropy(logits, targets, smoothing=0.0):
+    return CrossEntropyLS.apply(logits, targets, smoothing)
+
+
+# -------------------------------------------------- fused KD loss (SoftTarget, T=2)
+
+
+class SoftTargetKD(torch.autograd.Function):
+    """Hinton KD: KLDiv(log_softmax(s/T), softmax(t/T), batchmean) * T^2.
+
+    Reference utils.py:121-132 (SoftTarget). Gradient flows to student only.
+    """
+
+    @staticmethod
+    def forward(ctx, s_logits, t_logits, T):
+        ctx.T = T
+        if use_hip(s_logits):
+            loss, ps, pt = ext().kd_fwd(s_logits, t_logits, T)
+            ctx.save_for_backward(ps, pt)
+            return loss
+        sf, tf = s_logits.float() / T, t_logits.float() / T
+        ps = F.softmax(sf, dim=1)
+        pt = F.softmax(tf, dim=1)
+        logps = F.log_softmax(sf, dim=1)
+        logpt = F.log_softmax(tf, dim=1)
+        n = sf.shape[0]
+        loss = (pt * (logpt - logps)).sum() / n * (T * T)
+        ctx.save_for_backward(ps, pt)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ps, pt = ctx.saved_tensors
+        T = ctx.T
+        if use_hip(ps):
+            ds = ext().kd_bwd(ps, pt, T, dloss)
+            return ds, None, None
+        n = ps.shape[0]
+        ds = ((ps - pt) * (T / n) * dloss)
+        return ds, None, None
+
+
+def kd_loss(s_logits, t_logits, T=2.0):
+    return SoftTargetKD.apply(s_logits, t_logits, T)
+
+
+# ------------------------------------------------------------------------ max pool
+
+
+class MaxPoolNHWC(torch.autograd.Function):
+    """3x3/s2/p1-style max pool over NHWC (ImageNet-stem ResNets)."""
+
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        ctx.params = (kernel, stride, padding)
+        ctx.in_shape = x.shape
+        if use_hip(x):
+            y, idx = ext().maxpool_fwd(x, kernel, stride, padding)
+            ctx.save_for_backward(idx)
+            return y
+        xf = _to_nchw(x).float()
+        y, idx = F.max_pool2d(xf, kernel, stride, padding, return_indices=True)
+        ctx.save_for_backward(idx)
+        return _to_nhwc(y).to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        kernel, stride, padding = ctx.params
+        N, H, W, C = ctx.in_shape
+        dy = dy.contiguous()
+        if use_hip(dy):
+            return (ext().maxpool_bwd(dy, idx, H, W, kernel, stride, padding),
+                    None, None, None)
+        dyf = _to_nchw(dy).float()
+        # scatter-ADD (max_unpool overwrites on duplicate winners)
+        Nn, Cc = dyf.shape[0], dyf.shape[1]
+        dxf = dyf.new_zeros(Nn, Cc, H * W)
+        dxf.scatter_add_(2, idx.reshape(Nn, Cc, -1), dyf.reshape(Nn, Cc, -1))
+        dxf = dxf.reshape(Nn, Cc, H, W)
+        return _to_nhwc(dxf).to(dy.dtype), None, None, None
+
+
+def max_pool(x, kernel=3, stride=2, padding=1):
+    return MaxPoolNHWC.apply(x, kernel, stride, padding)
+
+
+# ------------------------------------------------------------------ top-k accuracy
+
+
+def accuracy(logits, targets, topk=(1,)):
+    """Top-k accuracy in percent (reference: timm.utils.accuracy, used
+    template.py:179-180, 267-268)."""
+    maxk = max(topk)
+    if logits.is_cuda and use_hip(logits):
+        counts = ext().topk_correct(logits, targets, maxk)  # (maxk,) cumulative
+        n = targets.shape[0]
+        return [counts[k - 1].item() * 100.0 / n for k in topk]
+    _, pred = logits.float().topk(maxk, dim=1)
+    correct = pred.eq(targets.view(-1, 1))
+    n = targets.shape[0]
+    return [correct[:, :k].any(dim=1).float().sum().item() * 100.0 / n for k in topk]

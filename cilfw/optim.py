@@ -1,0 +1,76 @@
+"""Fused SGD + cosine LR — the reference recipe as flat-buffer single-kernel ops.
+
+Reference: torch.optim.SGD(lr, momentum, weight_decay) + CosineAnnealingLR(T_max)
+(template.py:246-249, stepped per epoch :278). cilfw runs the whole update as ONE
+kernel over the DataParallelEngine's flat param/grad buffers (HIP kernel
+``sgd_step`` on GPU; two fused ATen ops on CPU):
+
+    g = grad + wd * p;  m = mu * m + g;  p -= lr * m
+"""
+
+import math
+
+import torch
+
+from .ops._backend import use_hip, ext
+
+
+class FlatSGD:
+    def __init__(self, engine, lr, momentum=0.9, weight_decay=5e-4):
+        self.engine = engine
+        self.lr = lr
+        self.base_lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self.momentum_buf = torch.zeros_like(engine.flat_grads)
+
+    @torch.no_grad()
+    def step(self):
+        p, g, m = (self.engine.flat_params, self.engine.flat_grads,
+                   self.momentum_buf)
+        if use_hip(p):
+            ext().sgd_step(p, g, m, self.lr, self.momentum, self.weight_decay)
+            return
+        # CPU reference: same math, fused at the ATen level
+        if self.weight_decay != 0:
+            g = g.add(p, alpha=self.weight_decay)
+        m.mul_(self.momentum).add_(g)
+        p.add_(m, alpha=-self.lr)
+
+    def zero_grad(self):
+        self.engine.zero_grad()
+
+    def state_dict(self):
+        return {"lr": self.lr, "base_lr": self.base_lr,
+                "momentum": self.momentum, "weight_decay": self.weight_decay,
+                "momentum_buf": self.momentum_buf}
+
+    def load_state_dict(self, sd):
+        self.lr = sd["lr"]
+        self.base_lr = sd["base_lr"]
+        self.momentum = sd["momentum"]
+        self.weight_decay = sd["weight_decay"]
+        self.momentum_buf.copy_(sd["momentum_buf"])
+
+
+class CosineLR:
+    """CosineAnnealingLR(T_max) equivalent, stepped once per epoch."""
+
+    def __init__(self, optimizer, t_max, eta_min=0.0):
+        self.opt = optimizer
+        self.t_max = t_max
+        self.eta_min = eta_min
+        self.epoch = 0
+
+    def step(self):
+        self.epoch += 1
+        self.opt.lr = self.eta_min + (self.opt.base_lr - self.eta_min) * \
+            (1 + math.cos(math.pi * min(self.epoch, self.t_max) / self.t_max)) / 2
+
+    def state_dict(self):
+        return {"epoch": self.epoch, "t_max": self.t_max, "eta_min": self.eta_min}
+
+    def load_state_dict(self, sd):
+        self.epoch = sd["epoch"]
+        self.t_max = sd["t_max"]
+        self.eta_min = sd["eta_min"]
