@@ -1,0 +1,283 @@
+"""HIP kernel numerics vs the CPU fp32 oracles (run on MI355X: pytest -m gpu).
+
+Every kernel is compared against the same op computed by the CPU reference
+implementation on the SAME bf16-quantized inputs (asymmetric random data — a
+transposed MFMA C-write or swapped operand cannot pass, guide §5.4 rule 16)."""
+
+import numpy as np
+import pytest
+import torch
+
+import cilfw.ops.functional as CF
+from cilfw.ops._backend import have_ext
+
+pytestmark = [pytest.mark.gpu,
+              pytest.mark.skipif(not torch.cuda.is_available(),
+                                 reason="needs GPU")]
+
+
+def _cmp(gpu, cpu, rtol=0.02, atol=0.02, what=""):
+    g = gpu.detach().float().cpu()
+    c = cpu.detach().float().cpu()
+    err = (g - c).abs()
+    scale = c.abs().max().clamp_min(1.0)
+    assert torch.isfinite(g).all(), f"{what}: non-finite GPU output"
+    ok = (err <= atol + rtol * scale).all()
+    assert ok, (f"{what}: max abs err {err.max().item():.4e} vs scale "
+                f"{scale.item():.3e}")
+
+
+def _pair(*shape, dtype=torch.bfloat16, seed=0, scale=1.0):
+    torch.manual_seed(seed)
+    t = (torch.randn(*shape) * scale).to(dtype)
+    return t.cuda(), t.cpu()
+
+
+def test_extension_loaded():
+    assert have_ext(), "cilfw/_hip_lib.so must be present on GPU boxes"
+
+
+# --------------------------------------------------------------------- conv
+
+@pytest.mark.parametrize("cin,cout,k,stride,hw,batch", [
+    (32, 64, 3, 1, 16, 4),    # fast A path (C%32==0)
+    (64, 128, 3, 2, 16, 4),   # strided
+    (3, 16, 3, 1, 32, 4),     # stem (generic path, ragged K)
+    (64, 64, 1, 1, 8, 4),     # 1x1
+    (64, 128, 1, 2, 8, 4),    # 1x1 stride-2 projection
+    (3, 64, 7, 2, 32, 2),     # imagenet stem
+    (48, 32, 3, 1, 8, 2),     # C not mult of 32 -> generic path
+])
+def test_conv_fwd_bwd(cin, cout, k, stride, hw, batch):
+    pad = k // 2
+    xg, xc = _pair(batch, hw, hw, cin, seed=1)
+    wg, wc = _pair(k, k, cin, cout, seed=2, scale=0.2)
+    wg32, wc32 = wg.float().requires_grad_(), wc.float().requires_grad_()
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+
+    yg = CF.conv2d(xg, wg32, stride, pad)
+    yc = CF.conv2d(xc, wc32, stride, pad)
+    _cmp(yg, yc, what=f"conv_fwd {cin}->{cout} k{k}s{stride}")
+
+    torch.manual_seed(3)
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    if cin >= 16:  # dx for the generic stem path is unused in practice
+        _cmp(xg.grad, xc.grad, what="conv_dx")
+    _cmp(wg32.grad, wc32.grad, rtol=0.03, atol=0.03, what="conv_dw")
+
+
+# ----------------------------------------------------------------------- bn
+
+@pytest.mark.parametrize("C,relu", [(16, False), (64, True), (256, True)])
+def test_bn_fwd_bwd(C, relu):
+    xg, xc = _pair(8, 6, 6, C, seed=4)
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    gg = (torch.rand(C) + 0.5)
+    bb = torch.randn(C)
+    ggg, ggc = gg.clone().cuda().requires_grad_(), gg.clone().requires_grad_()
+    bbg, bbc = bb.clone().cuda().requires_grad_(), bb.clone().requires_grad_()
+    rmg, rvg = torch.zeros(C).cuda(), torch.ones(C).cuda()
+    rmc, rvc = torch.zeros(C), torch.ones(C)
+
+    yg = CF.batchnorm_act(xg, ggg, bbg, rmg, rvg, training=True, relu=relu)
+    yc = CF.batchnorm_act(xc, ggc, bbc, rmc, rvc, training=True, relu=relu)
+    _cmp(yg, yc, what=f"bn_fwd C{C} relu{relu}")
+    _cmp(rmg, rmc, rtol=1e-3, atol=1e-4, what="bn running_mean")
+    _cmp(rvg, rvc, rtol=1e-3, atol=1e-4, what="bn running_var")
+
+    torch.manual_seed(5)
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, rtol=0.03, atol=0.03, what="bn_dx")
+    _cmp(ggg.grad, ggc.grad, rtol=0.02, atol=0.02, what="bn_dgamma")
+    _cmp(bbg.grad, bbc.grad, rtol=0.02, atol=0.02, what="bn_dbeta")
+
+
+def test_bn_eval():
+    C = 32
+    xg, xc = _pair(4, 5, 5, C, seed=6)
+    g, b = torch.rand(C) + 0.5, torch.randn(C)
+    rm, rv = torch.randn(C) * 0.1, torch.rand(C) + 0.5
+    yg = CF.batchnorm_act(xg, g.cuda(), b.cuda(), rm.cuda(), rv.cuda(),
+                          training=False)
+    yc = CF.batchnorm_act(xc, g, b, rm, rv, training=False)
+    _cmp(yg, yc, what="bn_eval")
+
+
+# -------------------------------------------------------------- elementwise
+
+def test_add_relu_downsample_gap():
+    ag, ac = _pair(4, 8, 8, 32, seed=7)
+    bg, bc = _pair(4, 8, 8, 32, seed=8)
+    ag, ac = ag.requires_grad_(), ac.requires_grad_()
+    yg, yc = CF.add_relu(ag, bg), CF.add_relu(ac, bc)
+    _cmp(yg, yc, what="add_relu")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(ag.grad, ac.grad, what="add_relu_bwd")
+
+    xg, xc = _pair(2, 8, 8, 16, seed=9)
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    yg, yc = CF.downsample_a(xg), CF.downsample_a(xc)
+    _cmp(yg, yc, what="downsample_a")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, what="downsample_a_bwd")
+
+    xg, xc = _pair(3, 8, 8, 64, seed=10)
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    yg, yc = CF.global_avg_pool(xg), CF.global_avg_pool(xc)
+    _cmp(yg, yc, what="gap")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, what="gap_bwd")
+
+
+def test_maxpool():
+    xg, xc = _pair(2, 16, 16, 32, seed=11)
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    yg, yc = CF.max_pool(xg, 3, 2, 1), CF.max_pool(xc, 3, 2, 1)
+    _cmp(yg, yc, what="maxpool")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, what="maxpool_bwd")
+
+
+# ------------------------------------------------------------------- linear
+
+@pytest.mark.parametrize("M,N,K", [(128, 100, 512), (37, 13, 64), (64, 10, 64)])
+def test_linear(M, N, K):
+    xg, xc = _pair(M, K, seed=12)
+    wg, wc = _pair(N, K, seed=13, scale=0.3)
+    wg32, wc32 = wg.float().requires_grad_(), wc.float().requires_grad_()
+    bg = torch.randn(N)
+    bg32, bc32 = bg.clone().cuda().requires_grad_(), bg.clone().requires_grad_()
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    yg = CF.linear(xg, wg32, bg32)
+    yc = CF.linear(xc, wc32, bc32)
+    _cmp(yg, yc, what=f"linear {M}x{N}x{K}")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, what="linear_dx")
+    _cmp(wg32.grad, wc32.grad, what="linear_dw")
+    _cmp(bg32.grad, bc32.grad, what="linear_db")
+
+
+# ------------------------------------------------------------------- losses
+
+@pytest.mark.parametrize("smooth", [0.0, 0.1])
+def test_ce(smooth):
+    torch.manual_seed(14)
+    logits = torch.randn(64, 110)
+    targets = torch.randint(0, 110, (64,))
+    lg = logits.cuda().requires_grad_()
+    lc = logits.clone().requires_grad_()
+    lossg = CF.cross_entropy(lg, targets.cuda(), smooth)
+    lossc = CF.cross_entropy(lc, targets, smooth)
+    _cmp(lossg, lossc, rtol=1e-4, atol=1e-5, what="ce_loss")
+    lossg.backward()
+    lossc.backward()
+    _cmp(lg.grad, lc.grad, rtol=1e-3, atol=1e-6, what="ce_grad")
+
+
+def test_kd():
+    torch.manual_seed(15)
+    s = torch.randn(32, 50)
+    t = torch.randn(32, 50)
+    sg = s.cuda().requires_grad_()
+    sc = s.clone().requires_grad_()
+    lg = CF.kd_loss(sg, t.cuda(), 2.0)
+    lc = CF.kd_loss(sc, t, 2.0)
+    _cmp(lg, lc, rtol=1e-4, atol=1e-5, what="kd_loss")
+    lg.backward()
+    lc.backward()
+    _cmp(sg.grad, sc.grad, rtol=1e-3, atol=1e-7, what="kd_grad")
+
+
+def test_sgd_step():
+    from cilfw.ops._backend import ext
+    torch.manual_seed(16)
+    n = 10007
+    p = torch.randn(n)
+    g = torch.randn(n)
+    m = torch.randn(n)
+    pg, gg, mg = p.clone().cuda(), g.clone().cuda(), m.clone().cuda()
+    ext().sgd_step(pg, gg, mg, 0.1, 0.9, 5e-4)
+    gc = g + 5e-4 * p
+    mc = 0.9 * m + gc
+    pc = p - 0.1 * mc
+    _cmp(pg, pc, rtol=1e-6, atol=1e-6, what="sgd_p")
+    _cmp(mg, mc, rtol=1e-6, atol=1e-6, what="sgd_m")
+
+
+def test_topk_accuracy():
+    torch.manual_seed(17)
+    logits = torch.randn(256, 100)
+    targets = torch.randint(0, 100, (256,))
+    got = CF.accuracy(logits.cuda(), targets.cuda(), topk=(1, 5))
+    want = CF.accuracy(logits, targets, topk=(1, 5))
+    assert got[0] == pytest.approx(want[0], abs=1e-6)
+    assert got[1] == pytest.approx(want[1], abs=1e-6)
+
+
+def test_herding_gpu_matches_cpu():
+    from cilfw.cil import herding_select
+    torch.manual_seed(18)
+    f = torch.randn(200, 64)
+    got = herding_select(f.cuda(), 50).cpu()
+    want = herding_select(f, 50)
+    # fp reduction order may swap near-ties; demand identical prefix objective
+    mu = f.mean(0)
+
+    def objective(order):
+        sel = f[order]
+        means = sel.cumsum(0) / torch.arange(1, len(order) + 1).unsqueeze(1)
+        return (means - mu).norm(dim=1)
+
+    og, ow = objective(got), objective(want)
+    assert torch.allclose(og, ow, rtol=1e-3, atol=1e-4), \
+        f"herding objective diverged: {(og - ow).abs().max()}"
+    # and the first picks should agree exactly
+    assert got[0].item() == want[0].item()
+
+
+# --------------------------------------------------------- whole-model parity
+
+def test_resnet20_step_matches_cpu():
+    """One fwd+bwd of the full model: GPU HIP path vs CPU oracle, same weights
+    (bf16 activations both sides; fp32 master params)."""
+    from cilfw.models import CilModel
+    torch.manual_seed(19)
+    mc = CilModel("resnet20", 32)
+    mc.prev_model_adaption(10)
+    mg = CilModel("resnet20", 32)
+    mg.prev_model_adaption(10)
+    mg.load_state_dict(mc.state_dict())
+    mg = mg.cuda()
+
+    x = torch.randn(8, 32, 32, 3).to(torch.bfloat16)
+    y = torch.randint(0, 10, (8,))
+    lg_logits, _ = mg(x.cuda())
+    lc_logits, _ = mc(x)
+    _cmp(lg_logits, lc_logits, rtol=0.05, atol=0.05, what="model logits")
+
+    lossg = CF.cross_entropy(lg_logits.float(), y.cuda())
+    lossc = CF.cross_entropy(lc_logits.float(), y)
+    _cmp(lossg, lossc, rtol=0.03, atol=0.03, what="model loss")
+    lossg.backward()
+    lossc.backward()
+    # compare a few representative grads
+    gstem_g = mg.backbone.conv_1_3x3.weight.grad
+    gstem_c = mc.backbone.conv_1_3x3.weight.grad
+    _cmp(gstem_g, gstem_c, rtol=0.08, atol=0.08, what="stem dw")
+    ghead_g = mg.fc.heads[0].weight.grad
+    ghead_c = mc.fc.heads[0].weight.grad
+    _cmp(ghead_g, ghead_c, rtol=0.05, atol=0.05, what="head dw")
