@@ -31,6 +31,38 @@ _lib.cilfw_sync.restype = c_i
 _lib.cilfw_error_string.restype = ctypes.c_char_p
 _lib.cilfw_error_string.argtypes = [c_i]
 
+# Explicit prototypes: ctypes cannot catch arity/type mismatches on its own and
+# a wrong call corrupts device memory (see gap_bwd incident, round 1).
+_PROTOS = {
+    "cilfw_conv2d_fwd": [c_vp] * 3 + [c_i] * 11 + [c_vp],
+    "cilfw_conv2d_bwd_data": [c_vp] * 3 + [c_i] * 11 + [c_vp],
+    "cilfw_conv2d_bwd_weight": [c_vp] * 3 + [c_i] * 11 + [c_vp],
+    "cilfw_bn_fwd": [c_vp] * 9 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
+    "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
+    "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
+    "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
+    "cilfw_downsample_a_fwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
+    "cilfw_downsample_a_bwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
+    "cilfw_gap_fwd": [c_vp] * 2 + [c_i] * 3 + [c_vp],
+    "cilfw_gap_bwd": [c_vp] * 2 + [c_i] * 3 + [c_vp],
+    "cilfw_maxpool_fwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
+    "cilfw_maxpool_bwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
+    "cilfw_linear_fwd": [c_vp] * 4 + [c_i] * 3 + [c_vp],
+    "cilfw_linear_dx": [c_vp] * 3 + [c_i] * 3 + [c_vp],
+    "cilfw_linear_dw": [c_vp] * 4 + [c_i] * 3 + [c_vp],
+    "cilfw_ce_fwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
+    "cilfw_ce_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
+    "cilfw_kd_fwd": [c_vp] * 5 + [c_i, c_i, c_f, c_vp],
+    "cilfw_kd_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
+    "cilfw_sgd_step": [c_vp] * 3 + [c_l, c_f, c_f, c_f, c_vp],
+    "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
+    "cilfw_herding_select": [c_vp] * 3 + [c_i] * 3 + [c_vp],
+}
+for _name, _args in _PROTOS.items():
+    _fn = getattr(_lib, _name)
+    _fn.argtypes = _args
+    _fn.restype = None
+
 
 def _stream():
     return c_vp(torch.cuda.current_stream().cuda_stream)
@@ -193,8 +225,8 @@ def gap_fwd(x):
 def gap_bwd(dy, H, W_):
     N, C = dy.shape
     dx = torch.empty(N, H, W_, C, dtype=dy.dtype, device=dy.device)
-    _lib.cilfw_gap_bwd(_ptr(dy), _ptr(dx), c_i(H * W_), c_i(C),
-                       c_l(N * H * W_ * C), _stream())
+    _lib.cilfw_gap_bwd(_ptr(dy), _ptr(dx), c_i(N), c_i(H * W_), c_i(C),
+                       _stream())
     _check("gap_bwd")
     return dx
 
