@@ -97,6 +97,7 @@ def conv2d_fwd(x, w, stride, pad):
     N, H, W_, C = x.shape
     R, S, Cw, K = w.shape
     assert Cw == C
+    assert K % 8 == 0, "conv kernels vector-stage over output channels (K%8==0)"
     Ho = (H + 2 * pad - R) // stride + 1
     Wo = (W_ + 2 * pad - S) // stride + 1
     y = torch.empty(N, Ho, Wo, K, dtype=torch.bfloat16, device=x.device)
@@ -141,6 +142,7 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
            relu):
     _bf16(x, "bn_fwd.x")
     C = x.shape[-1]
+    assert C % 8 == 0, "bn kernels vectorize over channels (C%8==0)"
     M = x.numel() // C
     y = torch.empty_like(x)
     mean = torch.empty(C, dtype=torch.float32, device=x.device)

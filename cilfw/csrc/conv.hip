@@ -119,13 +119,21 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
       areg0 = *(int4*)&tmp[0];
       areg1 = *(int4*)&tmp[8];
     }
-    // ---- B: Bs[n][kk] = w[(k0+kk)*K + ko0+n]; thread: n = t&63, kk = t>>6 + 4i
-    const int bn = t & 63, bk0 = t >> 6;
+    // ---- B: Bs[n][kk] = w[(k0+kk)*K + ko0+n]
+    // thread: kk = t>>3 (0..31), ng = t&7 (8 couts per b128 load; K % 8 == 0)
+    const int bkk = t >> 3, bng = t & 7;
+    {
+      int k = k0 + bkk;
+      if (k < CRS && ko0 + bng * 8 + 8 <= g.K) {
+        *(int4*)breg = *(const int4*)&w[(long)k * g.K + ko0 + bng * 8];
+      } else if (k < CRS && ko0 + bng * 8 < g.K) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int kk = bk0 + i * 4;
-      int k = k0 + kk;
-      breg[i] = (k < CRS && ko0 + bn < g.K) ? w[(long)k * g.K + ko0 + bn] : 0;
+        for (int j = 0; j < 8; ++j)
+          breg[j] = (ko0 + bng * 8 + j < g.K)
+                        ? w[(long)k * g.K + ko0 + bng * 8 + j] : 0;
+      } else {
+        *(int4*)breg = int4{0, 0, 0, 0};
+      }
     }
   };
 
@@ -134,9 +142,9 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     bf16_t* Bs = &lds[BS_OFF(buf)];
     *(int4*)&As[arow * LP + ahalf * 16] = areg0;
     *(int4*)&As[arow * LP + ahalf * 16 + 8] = areg1;
-    const int bn = t & 63, bk0 = t >> 6;
+    const int bkk = t >> 3, bng = t & 7;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) Bs[bn * LP + bk0 + i * 4] = breg[i];
+    for (int j = 0; j < 8; ++j) Bs[(bng * 8 + j) * LP + bkk] = breg[j];
   };
 
   stage_to_regs(0);
@@ -244,18 +252,18 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       areg0 = *(int4*)&tmp[0];
       areg1 = *(int4*)&tmp[8];
     }
-    // B: Bs[c][kk] = w[(rs*C + c0+c)*K + kc]; thread: kk = t&31, c = t>>5 + 8i
-    const int bkk = t & 31, bc0 = t >> 5;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int c = bc0 + i * 8;
+    // B: Bs[c][kk] = w[(rs*C + c0+c)*K + kc]
+    // thread: c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0 so
+    // an 8-chunk never crosses an (r,s) boundary)
+    const int bc = t >> 2, bkk = (t & 3) * 8;
+    {
       int k = k0 + bkk;
-      bf16_t v = 0;
-      if (k < RSK && c0 + c < g.C) {
+      if (k < RSK && c0 + bc < g.C) {
         int rs = k / g.K, kc = k - rs * g.K;
-        v = w[((long)rs * g.C + c0 + c) * g.K + kc];
+        *(int4*)breg = *(const int4*)&w[((long)rs * g.C + c0 + bc) * g.K + kc];
+      } else {
+        *(int4*)breg = int4{0, 0, 0, 0};
       }
-      breg[i] = v;
     }
   };
 
@@ -264,9 +272,8 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     bf16_t* Bs = &lds[BS_OFF(buf)];
     *(int4*)&As[arow * LP + ahalf * 16] = areg0;
     *(int4*)&As[arow * LP + ahalf * 16 + 8] = areg1;
-    const int bkk = t & 31, bc0 = t >> 5;
-#pragma unroll
-    for (int i = 0; i < 8; ++i) Bs[(bc0 + i * 8) * LP + bkk] = breg[i];
+    const int bc = t >> 2, bkk = (t & 3) * 8;
+    *(int4*)&Bs[bc * LP + bkk] = *(int4*)breg;
   };
 
   stage_to_regs(0);
@@ -310,7 +317,7 @@ __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
                               const bf16_t* __restrict__ x,
                               float* __restrict__ dw, ConvGeom g, int M,
-                              int CRS, int slice_len) {
+                              int CRS, int slice_len, int fast_a) {
   __shared__ bf16_t lds[WLDS_ELEMS];
   const int rs0 = blockIdx.x * WBM;   // CRS rows
   const int ko0 = blockIdx.y * BN;    // Kout cols
@@ -319,15 +326,16 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
 
-  // A row (CRS index) decomposition for this thread's 2 rows (t&63 + 64*i)? No:
-  // staging mapping: row = t & 63, mm = t >> 6 (+4i)
-  const int arow = t & 63, amm0 = t >> 6;
-  int r_ = 0, s_ = 0, c_ = 0;
-  bool arow_ok = rs0 + arow < CRS;
-  if (arow_ok) {
-    int k = rs0 + arow;
-    int rs = k / g.C;
-    c_ = k - rs * g.C;
+  // staging mapping: mm = t>>3 (0..31 reduction cols), grp = t&7 (8-row group)
+  const int amm = t >> 3, agrp = t & 7;
+  // A rows rs0 + agrp*8 + j (j=0..7): when C % 8 == 0 the 8-row group stays in
+  // ONE (r,s) with contiguous c -> one b128 gather per m (fast path)
+  int r_ = 0, s_ = 0, cbase_ = 0;
+  const int rowb = rs0 + agrp * 8;
+  bool agrp_ok = rowb < CRS;
+  if (agrp_ok) {
+    int rs = rowb / g.C;
+    cbase_ = rowb - rs * g.C;
     r_ = rs / g.S;
     s_ = rs - r_ * g.S;
   }
@@ -338,34 +346,56 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  bf16_t areg[8], breg[8];
+  __align__(16) bf16_t areg[8];
+  __align__(16) bf16_t breg[8];
   const int HoWo = g.Ho * g.Wo;
 
   auto stage_to_regs = [&](int m0) {
-    // A[arow][mm] = X[n, ho*st-pad+r_, wo*st-pad+s_, c_] for m = m0+mm
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int mm = amm0 + i * 4;
-      int m = m0 + mm;
-      bf16_t v = 0;
-      if (arow_ok && m < me) {
-        int n = m / HoWo;
-        int rem = m - n * HoWo;
-        int ho = rem / g.Wo, wo = rem - ho * g.Wo;
-        int hi = ho * g.stride - g.pad + r_;
-        int wi = wo * g.stride - g.pad + s_;
-        if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-          v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c_];
-      }
-      areg[i] = v;
+    const int m = m0 + amm;
+    int n = 0, ho = 0, wo = 0;
+    const bool m_ok = m < me;
+    if (m_ok) {
+      n = m / HoWo;
+      int rem = m - n * HoWo;
+      ho = rem / g.Wo;
+      wo = rem - ho * g.Wo;
     }
-    // Bt[kout][mm] = dY[(m0+mm)*K + ko0+kout]; thread: kout = t&63, mm = t>>6+4i
+    // ---- A: 8 rows (rsc) x 1 col (m): b128 gather from x
+    if (fast_a) {
+      int hi = ho * g.stride - g.pad + r_;
+      int wi = wo * g.stride - g.pad + s_;
+      if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
+        *(int4*)areg = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
+                                        * g.C + cbase_];
+      } else {
+        *(int4*)areg = int4{0, 0, 0, 0};
+      }
+    } else {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int mm = amm0 + i * 4;
-      int m = m0 + mm;
-      breg[i] = (m < me && ko0 + arow < g.K)
-                    ? dy[(long)m * g.K + ko0 + arow] : 0;
+      for (int j = 0; j < 8; ++j) {
+        int row = rowb + j;
+        bf16_t v = 0;
+        if (row < CRS && m_ok) {
+          int rs = row / g.C, c = row - rs * g.C;
+          int r = rs / g.S, s = rs - r * g.S;
+          int hi = ho * g.stride - g.pad + r;
+          int wi = wo * g.stride - g.pad + s;
+          if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+            v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c];
+        }
+        areg[j] = v;
+      }
+    }
+    // ---- B: 8 kouts x 1 col (m): b128 from dy (K % 8 == 0)
+    if (m_ok && ko0 + agrp * 8 + 8 <= g.K) {
+      *(int4*)breg = *(const int4*)&dy[(long)m * g.K + ko0 + agrp * 8];
+    } else if (m_ok) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        breg[j] = (ko0 + agrp * 8 + j < g.K)
+                      ? dy[(long)m * g.K + ko0 + agrp * 8 + j] : 0;
+    } else {
+      *(int4*)breg = int4{0, 0, 0, 0};
     }
   };
 
@@ -373,9 +403,9 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     bf16_t* As = &lds[WAS_OFF(buf)];
     bf16_t* Bs = &lds[WBS_OFF(buf)];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) As[arow * LP + amm0 + i * 4] = areg[i];
+    for (int j = 0; j < 8; ++j) As[(agrp * 8 + j) * LP + amm] = areg[j];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) Bs[arow * LP + amm0 + i * 4] = breg[i];
+    for (int j = 0; j < 8; ++j) Bs[(agrp * 8 + j) * LP + amm] = breg[j];
   };
 
   const int nk = cdiv_i(me - ms, BK);
@@ -465,13 +495,15 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, void* dw,
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
-  hipMemsetAsync(dw, 0, (size_t)CRS * K * sizeof(float), (hipStream_t)stream);
+  (void)hipMemsetAsync(dw, 0, (size_t)CRS * K * sizeof(float),
+                       (hipStream_t)stream);
   int slice_len = 4096;
   int nslices = cdiv(M, slice_len);
+  int fast_a = (C % 8 == 0);
   dim3 grid(cdiv(CRS, WBM), cdiv(K, BN), nslices);
   hipLaunchKernelGGL(conv2d_bwd_weight_kernel, grid, dim3(NTHREADS), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)x,
-                     (float*)dw, g, M, CRS, slice_len);
+                     (float*)dw, g, M, CRS, slice_len, fast_a);
 }
 
 }  // extern "C"

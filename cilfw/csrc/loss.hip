@@ -285,7 +285,7 @@ extern "C" {
 
 void cilfw_ce_fwd(const void* logits, const void* targets, void* probs,
                   void* loss, int M, int C, float smooth, void* stream) {
-  hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
+  (void)hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
   hipLaunchKernelGGL(ce_fwd_kernel, dim3(M), dim3(NT), 0,
                      (hipStream_t)stream, (const float*)logits,
                      (const long*)targets, (float*)probs, (float*)loss, M, C,
@@ -303,7 +303,7 @@ void cilfw_ce_bwd(const void* probs, const void* targets, const void* dloss,
 
 void cilfw_kd_fwd(const void* s, const void* t, void* ps, void* pt,
                   void* loss, int M, int C, float T, void* stream) {
-  hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
+  (void)hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
   hipLaunchKernelGGL(kd_fwd_kernel, dim3(M), dim3(NT), 0,
                      (hipStream_t)stream, (const float*)s, (const float*)t,
                      (float*)ps, (float*)pt, (float*)loss, M, C, T);
@@ -328,7 +328,7 @@ void cilfw_sgd_step(void* p, const void* g, void* m, long n, float lr,
 
 void cilfw_topk_correct(const void* logits, const void* targets, void* counts,
                         int M, int C, int maxk, void* stream) {
-  hipMemsetAsync(counts, 0, maxk * sizeof(long), (hipStream_t)stream);
+  (void)hipMemsetAsync(counts, 0, maxk * sizeof(long), (hipStream_t)stream);
   hipLaunchKernelGGL(topk_kernel, dim3(M), dim3(NT), 0, (hipStream_t)stream,
                      (const float*)logits, (const long*)targets,
                      (long*)counts, M, C, maxk);

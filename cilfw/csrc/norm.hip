@@ -11,33 +11,58 @@
 // pass 1: per-channel sum & sumsq via per-block partials + fp32 atomics.
 // x viewed as (M, C); grid.x covers C/64, grid.y covers row-chunks.
 
+// 256 threads = 32 row-lanes x 8 channel-groups of 8 (b128 loads); the block
+// covers a 64-channel slice (C % 8 == 0 in all cilfw models; C < 64 handled by
+// per-group bounds) over rows_per_blk rows, partials reduced in LDS, then one
+// fp32 atomic per (channel, quantity) per block.
 __global__ __launch_bounds__(NT)
 void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ sum,
                     float* __restrict__ sumsq, long M, int C, int rows_per_blk) {
-  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  const int rlane = threadIdx.x >> 6;  // 0..3
-  const bool active = c < C;  // never early-return before __syncthreads
-  long r0 = (long)blockIdx.y * rows_per_blk + rlane;
+  const int cg = threadIdx.x & 7;          // channel group (8 ch)
+  const int rl = threadIdx.x >> 3;         // row lane 0..31
+  const int c8 = blockIdx.x * 64 + cg * 8;
+  const bool active = c8 + 8 <= C || c8 < C;
+  long r0 = (long)blockIdx.y * rows_per_blk + rl;
   long r1 = min((long)(blockIdx.y + 1) * rows_per_blk, M);
-  float s = 0.f, q = 0.f;
-  if (active)
-    for (long r = r0; r < r1; r += 4) {
-      float v = bf2f(x[r * C + c]);
-      s += v;
-      q += v * v;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (c8 + 8 <= C) {
+    for (long r = r0; r < r1; r += 32) {
+      int4 v = *(const int4*)&x[r * C + c8];
+      const bf16_t* e = (const bf16_t*)&v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(e[j]);
+        s[j] += f;
+        q[j] += f * f;
+      }
     }
-  // reduce the 4 row-lanes of this channel through LDS
-  __shared__ float red[2][4][64];
-  red[0][rlane][threadIdx.x & 63] = s;
-  red[1][rlane][threadIdx.x & 63] = q;
+  } else if (active) {  // ragged channel tail
+    for (long r = r0; r < r1; r += 32)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (c8 + j < C) {
+          float f = bf2f(x[r * C + c8 + j]);
+          s[j] += f;
+          q[j] += f * f;
+        }
+  }
+  __shared__ float red[2][32][64];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][rl][cg * 8 + j] = s[j];
+    red[1][rl][cg * 8 + j] = q[j];
+  }
   __syncthreads();
-  if (rlane == 0 && active) {
-    s = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
-        red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
-    q = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
-        red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-    atomicAdd(&sum[c], s);
-    atomicAdd(&sumsq[c], q);
+  // 128 threads each own one (quantity, channel): serial sum over 32 row lanes
+  const int qi = threadIdx.x >> 7;         // 0..1 (threads 0..255 -> 2x128)
+  const int cc = threadIdx.x & 127;
+  if (cc < 64 && blockIdx.x * 64 + cc < C) {
+    float acc = 0.f;
+#pragma unroll 8
+    for (int r = 0; r < 32; ++r) acc += red[qi][r][cc];
+    atomicAdd(qi == 0 ? &sum[blockIdx.x * 64 + cc]
+                      : &sumsq[blockIdx.x * 64 + cc], acc);
   }
 }
 
@@ -117,33 +142,72 @@ void bn_bwd_sums_kernel(const bf16_t* __restrict__ dy,
                         const float* __restrict__ invstd,
                         float* __restrict__ dgamma, float* __restrict__ dbeta,
                         long M, int C, int rows_per_blk, int relu) {
-  const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-  const int rlane = threadIdx.x >> 6;
-  const bool active = c < C;
-  const float mu = active ? mean[c] : 0.f, is = active ? invstd[c] : 0.f;
-  long r0 = (long)blockIdx.y * rows_per_blk + rlane;
+  const int cg = threadIdx.x & 7;
+  const int rl = threadIdx.x >> 3;
+  const int c8 = blockIdx.x * 64 + cg * 8;
+  float mu[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c8 + j;
+    mu[j] = (c < C) ? mean[c] : 0.f;
+    is[j] = (c < C) ? invstd[c] : 0.f;
+  }
+  long r0 = (long)blockIdx.y * rows_per_blk + rl;
   long r1 = min((long)(blockIdx.y + 1) * rows_per_blk, M);
-  float sg = 0.f, sb = 0.f;
-  if (active)
-    for (long r = r0; r < r1; r += 4) {
-      long i = r * C + c;
-      float g = bf2f(dy[i]);
-      if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
-      float xhat = (bf2f(x[i]) - mu) * is;
-      sb += g;
-      sg += g * xhat;
+  float sg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float sb[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (c8 + 8 <= C) {
+    for (long r = r0; r < r1; r += 32) {
+      long i = r * C + c8;
+      int4 dv = *(const int4*)&dy[i];
+      int4 xv = *(const int4*)&x[i];
+      const bf16_t* de = (const bf16_t*)&dv;
+      const bf16_t* xe = (const bf16_t*)&xv;
+      if (relu) {
+        int4 yv = *(const int4*)&y[i];
+        const bf16_t* ye = (const bf16_t*)&yv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = (bf2f(ye[j]) > 0.f) ? bf2f(de[j]) : 0.f;
+          sb[j] += g;
+          sg[j] += g * (bf2f(xe[j]) - mu[j]) * is[j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = bf2f(de[j]);
+          sb[j] += g;
+          sg[j] += g * (bf2f(xe[j]) - mu[j]) * is[j];
+        }
+      }
     }
-  __shared__ float red[2][4][64];
-  red[0][rlane][threadIdx.x & 63] = sg;
-  red[1][rlane][threadIdx.x & 63] = sb;
+  } else if (c8 < C) {
+    for (long r = r0; r < r1; r += 32)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (c8 + j < C) {
+          long i = r * C + c8 + j;
+          float g = bf2f(dy[i]);
+          if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
+          sb[j] += g;
+          sg[j] += g * (bf2f(x[i]) - mu[j]) * is[j];
+        }
+  }
+  __shared__ float red[2][32][64];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][rl][cg * 8 + j] = sg[j];
+    red[1][rl][cg * 8 + j] = sb[j];
+  }
   __syncthreads();
-  if (rlane == 0 && active) {
-    sg = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
-         red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
-    sb = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
-         red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-    atomicAdd(&dgamma[c], sg);
-    atomicAdd(&dbeta[c], sb);
+  const int qi = threadIdx.x >> 7;
+  const int cc = threadIdx.x & 127;
+  if (cc < 64 && blockIdx.x * 64 + cc < C) {
+    float acc = 0.f;
+#pragma unroll 8
+    for (int r = 0; r < 32; ++r) acc += red[qi][r][cc];
+    atomicAdd(qi == 0 ? &dgamma[blockIdx.x * 64 + cc]
+                      : &dbeta[blockIdx.x * 64 + cc], acc);
   }
 }
 
@@ -359,10 +423,10 @@ void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
                   void* stream) {
   hipStream_t st = (hipStream_t)stream;
   if (training) {
-    hipMemsetAsync(scratch_sums, 0, 2 * C * sizeof(float), st);
+    (void)hipMemsetAsync(scratch_sums, 0, 2 * C * sizeof(float), st);
     float* sum = (float*)scratch_sums;
     float* sumsq = sum + C;
-    int rows_per_blk = (int)min((long)4096, max((long)256, M / 64));
+    int rows_per_blk = 256;
     dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
     hipLaunchKernelGGL(bn_sums_kernel, grid, dim3(NT), 0, st,
                        (const bf16_t*)x, sum, sumsq, M, C, rows_per_blk);
@@ -389,9 +453,9 @@ void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                   void* dgamma, void* dbeta, long M, int C, int relu,
                   int training, void* stream) {
   hipStream_t st = (hipStream_t)stream;
-  hipMemsetAsync(dgamma, 0, C * sizeof(float), st);
-  hipMemsetAsync(dbeta, 0, C * sizeof(float), st);
-  int rows_per_blk = (int)min((long)4096, max((long)256, M / 64));
+  (void)hipMemsetAsync(dgamma, 0, C * sizeof(float), st);
+  (void)hipMemsetAsync(dbeta, 0, C * sizeof(float), st);
+  int rows_per_blk = 256;
   dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
   hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
                      (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
