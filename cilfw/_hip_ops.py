@@ -34,9 +34,9 @@ _lib.cilfw_error_string.argtypes = [c_i]
 # Explicit prototypes: ctypes cannot catch arity/type mismatches on its own and
 # a wrong call corrupts device memory (see gap_bwd incident, round 1).
 _PROTOS = {
-    "cilfw_conv2d_fwd": [c_vp] * 3 + [c_i] * 11 + [c_vp],
-    "cilfw_conv2d_bwd_data": [c_vp] * 3 + [c_i] * 11 + [c_vp],
-    "cilfw_conv2d_bwd_weight": [c_vp] * 3 + [c_i] * 11 + [c_vp],
+    "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_bwd_weight": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_bn_fwd": [c_vp] * 9 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
     "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
@@ -62,6 +62,13 @@ for _name, _args in _PROTOS.items():
     _fn = getattr(_lib, _name)
     _fn.argtypes = _args
     _fn.restype = None
+
+for _name, _n in [("cilfw_conv2d_fwd_ksplit", 7),
+                  ("cilfw_conv2d_bwd_data_ksplit", 7),
+                  ("cilfw_conv2d_bwd_weight_nslices", 7)]:
+    _fn = getattr(_lib, _name)
+    _fn.argtypes = [c_i] * _n
+    _fn.restype = c_i
 
 
 def _stream():
@@ -101,9 +108,13 @@ def conv2d_fwd(x, w, stride, pad):
     Ho = (H + 2 * pad - R) // stride + 1
     Wo = (W_ + 2 * pad - S) // stride + 1
     y = torch.empty(N, Ho, Wo, K, dtype=torch.bfloat16, device=x.device)
-    _lib.cilfw_conv2d_fwd(_ptr(x), _ptr(w), _ptr(y), c_i(N), c_i(H), c_i(W_),
-                          c_i(C), c_i(K), c_i(R), c_i(S), c_i(stride),
-                          c_i(pad), c_i(Ho), c_i(Wo), _stream())
+    ks = _lib.cilfw_conv2d_fwd_ksplit(N, C, K, R, S, Ho, Wo)
+    ws = (torch.empty(ks * N * Ho * Wo * K, dtype=torch.float32,
+                      device=x.device) if ks > 1 else None)
+    _lib.cilfw_conv2d_fwd(_ptr(x), _ptr(w), _ptr(y), _ptr(ws), c_i(N),
+                          c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
+                          c_i(stride), c_i(pad), c_i(Ho), c_i(Wo), c_i(ks),
+                          _stream())
     _check("conv2d_fwd")
     return y
 
@@ -114,10 +125,13 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
     R, S, C, Kw = w.shape
     assert Kw == K
     dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
-    _lib.cilfw_conv2d_bwd_data(_ptr(dy), _ptr(w), _ptr(dx), c_i(N), c_i(H),
-                               c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
-                               c_i(stride), c_i(pad), c_i(Ho), c_i(Wo),
-                               _stream())
+    ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H, W_, C, K, R, S)
+    ws = (torch.empty(ks * N * H * W_ * C, dtype=torch.float32,
+                      device=dy.device) if ks > 1 else None)
+    _lib.cilfw_conv2d_bwd_data(_ptr(dy), _ptr(w), _ptr(dx), _ptr(ws), c_i(N),
+                               c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R),
+                               c_i(S), c_i(stride), c_i(pad), c_i(Ho),
+                               c_i(Wo), c_i(ks), _stream())
     _check("conv2d_bwd_data")
     return dx
 
@@ -128,10 +142,13 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
     dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
-    _lib.cilfw_conv2d_bwd_weight(_ptr(dy), _ptr(x), _ptr(dw), c_i(N), c_i(H),
-                                 c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
-                                 c_i(stride), c_i(pad), c_i(Ho), c_i(Wo),
-                                 _stream())
+    ns = _lib.cilfw_conv2d_bwd_weight_nslices(N, C, K, R, S, Ho, Wo)
+    ws = torch.empty(ns * R * S * C * K, dtype=torch.float32,
+                     device=dy.device)
+    _lib.cilfw_conv2d_bwd_weight(_ptr(dy), _ptr(x), _ptr(dw), _ptr(ws),
+                                 c_i(N), c_i(H), c_i(W_), c_i(C), c_i(K),
+                                 c_i(R), c_i(S), c_i(stride), c_i(pad),
+                                 c_i(Ho), c_i(Wo), c_i(ns), _stream())
     _check("conv2d_bwd_weight")
     return dw
 

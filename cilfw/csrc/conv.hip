@@ -7,8 +7,12 @@
 // Tile: 128(M) x 64(Kout) x 32(K-step), 4 waves as 2x2, each wave 64x32 via
 // mfma_f32_16x16x32_bf16 (M_rep=4, N_rep=2), fp32 accumulators, double-buffered
 // LDS with +16B row padding (bank-conflict-free ds_read_b128).
-// Fast A-staging path when C % 32 == 0 (every K-chunk lies inside one (r,s) —
-// 16B vector loads); generic per-element gather otherwise (stem convs).
+//
+// Late layers have small M (batch x 4x4 spatial) and huge K-dim (C*R*S up to
+// 4608): too few workgroups to fill 256 CUs. When (M-blocks x K-blocks) is
+// small the launcher SPLITS the K loop over blockIdx.z; each slice writes an
+// fp32 partial slab and a reduce kernel sums slabs -> bf16 (no atomics).
+// Backward-weight always uses the slab path (it is a huge-reduction GEMM).
 
 #include "common.h"
 
@@ -18,8 +22,6 @@
 #define LP (BK + 8)          // LDS row pitch in bf16 elements (+16B pad)
 #define NTHREADS 256
 
-// one __shared__ object only (see guide §5 trap 4a)
-// As: 2 * BM * LP, Bs: 2 * BN * LP
 #define AS_OFF(buf) ((buf) * BM * LP)
 #define BS_OFF(buf) (2 * BM * LP + (buf) * BN * LP)
 #define LDS_ELEMS (2 * BM * LP + 2 * BN * LP)
@@ -32,7 +34,7 @@ struct ConvGeom {
 DEV void mfma_tile(const bf16_t* lds, int a_off, int b_off, int wr, int wc,
                    f32x4 acc[4][2]) {
   FragIdx fi = frag_idx();
-  const int kb = fi.quad * 8;  // this lane's k-offset within the 32-chunk
+  const int kb = fi.quad * 8;
 #pragma unroll
   for (int mr = 0; mr < 4; ++mr) {
     int row = wr * 64 + mr * 16 + fi.half;
@@ -47,20 +49,69 @@ DEV void mfma_tile(const bf16_t* lds, int a_off, int b_off, int wr, int wc,
   }
 }
 
+// sum ksplit fp32 slabs [ns][len] -> bf16 out[len]
+__global__ __launch_bounds__(NTHREADS)
+void reduce_slabs_bf16_kernel(const float* __restrict__ ws,
+                              bf16_t* __restrict__ out, int ns, long len) {
+  long i0 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 4;
+  if (i0 >= len) return;
+  if (i0 + 4 <= len) {
+    float4 acc = *(const float4*)&ws[i0];
+    for (int s = 1; s < ns; ++s) {
+      float4 v = *(const float4*)&ws[(long)s * len + i0];
+      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    }
+    bf16_t o[4] = {f2bf(acc.x), f2bf(acc.y), f2bf(acc.z), f2bf(acc.w)};
+    *(int2*)&out[i0] = *(int2*)o;
+  } else {
+    for (long i = i0; i < len; ++i) {
+      float a = ws[i];
+      for (int s = 1; s < ns; ++s) a += ws[(long)s * len + i];
+      out[i] = f2bf(a);
+    }
+  }
+}
+
+// sum ksplit fp32 slabs -> fp32 out (weight grads)
+__global__ __launch_bounds__(NTHREADS)
+void reduce_slabs_f32_kernel(const float* __restrict__ ws,
+                             float* __restrict__ out, int ns, long len) {
+  long i0 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 4;
+  if (i0 >= len) return;
+  if (i0 + 4 <= len) {
+    float4 acc = *(const float4*)&ws[i0];
+    for (int s = 1; s < ns; ++s) {
+      float4 v = *(const float4*)&ws[(long)s * len + i0];
+      acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+    }
+    *(float4*)&out[i0] = acc;
+  } else {
+    for (long i = i0; i < len; ++i) {
+      float a = ws[i];
+      for (int s = 1; s < ns; ++s) a += ws[(long)s * len + i];
+      out[i] = a;
+    }
+  }
+}
+
 // ============================== forward ==============================
 
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        const bf16_t* __restrict__ w,
-                       bf16_t* __restrict__ y, ConvGeom g, int M, int CRS,
-                       int nk, int fast_a) {
+                       bf16_t* __restrict__ y, float* __restrict__ ws,
+                       ConvGeom g, int M, int CRS, int nk, int fast_a,
+                       int ksplit) {
   __shared__ bf16_t lds[LDS_ELEMS];
   const int m0 = blockIdx.x * BM;
   const int ko0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
+  // this z-slice's K-step range
+  const int steps = (nk + ksplit - 1) / ksplit;
+  const int kt0 = blockIdx.z * steps;
+  const int kt1 = min(kt0 + steps, nk);
 
-  // per-thread A-staging coords: 2 threads per row, 16 elems each
   const int arow = t >> 1, ahalf = t & 1;
   int m = m0 + arow;
   int an = 0, aho = 0, awo = 0;
@@ -71,7 +122,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     aho = rem / g.Wo;
     awo = rem - aho * g.Wo;
   }
-  const int ahb = aho * g.stride - g.pad;  // base input coords
+  const int ahb = aho * g.stride - g.pad;
   const int awb = awo * g.stride - g.pad;
 
   f32x4 acc[4][2];
@@ -80,13 +131,11 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  // staging registers
   int4 areg0, areg1;
-  bf16_t breg[8];
+  __align__(16) bf16_t breg[8];
 
   auto stage_to_regs = [&](int kt) {
     const int k0 = kt * BK;
-    // ---- A: 16 elems for (arow, ahalf) ----
     if (fast_a) {
       int k = k0 + ahalf * 16;
       int rs = k / g.C, c0 = k - rs * g.C;
@@ -102,7 +151,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         areg1 = int4{0, 0, 0, 0};
       }
     } else {
-      bf16_t tmp[16];
+      __align__(16) bf16_t tmp[16];
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         int k = k0 + ahalf * 16 + j;
@@ -119,8 +168,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
       areg0 = *(int4*)&tmp[0];
       areg1 = *(int4*)&tmp[8];
     }
-    // ---- B: Bs[n][kk] = w[(k0+kk)*K + ko0+n]
-    // thread: kk = t>>3 (0..31), ng = t&7 (8 couts per b128 load; K % 8 == 0)
+    // B: thread: kk = t>>3 (0..31), ng = t&7 (8 couts; K % 8 == 0)
     const int bkk = t >> 3, bng = t & 7;
     {
       int k = k0 + bkk;
@@ -147,21 +195,22 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     for (int j = 0; j < 8; ++j) Bs[(bng * 8 + j) * LP + bkk] = breg[j];
   };
 
-  stage_to_regs(0);
-  regs_to_lds(0);
+  if (kt0 < kt1) {
+    stage_to_regs(kt0);
+    regs_to_lds(0);
+  }
   __syncthreads();
-  for (int kt = 0; kt < nk; ++kt) {
-    int cur = kt & 1;
-    if (kt + 1 < nk) stage_to_regs(kt + 1);      // global loads overlap MFMA
+  for (int kt = kt0; kt < kt1; ++kt) {
+    int cur = (kt - kt0) & 1;
+    if (kt + 1 < kt1) stage_to_regs(kt + 1);
     mfma_tile(lds, AS_OFF(cur), BS_OFF(cur), wr, wc, acc);
     __syncthreads();
-    if (kt + 1 < nk) {
+    if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
       __syncthreads();
     }
   }
 
-  // epilogue: C/D map col=lane&15, row=(lane>>4)*4+reg
   FragIdx fi = frag_idx();
 #pragma unroll
   for (int mr = 0; mr < 4; ++mr)
@@ -171,8 +220,12 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + mr * 16 + fi.quad * 4 + r;
         int col = ko0 + wc * 32 + nr * 16 + fi.half;
-        if (row < M && col < g.K)
-          y[(long)row * g.K + col] = f2bf(acc[mr][nr][r]);
+        if (row < M && col < g.K) {
+          if (ksplit > 1)
+            ws[((long)blockIdx.z * M + row) * g.K + col] = acc[mr][nr][r];
+          else
+            y[(long)row * g.K + col] = f2bf(acc[mr][nr][r]);
+        }
       }
 }
 
@@ -182,13 +235,17 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
                             const bf16_t* __restrict__ w,
-                            bf16_t* __restrict__ dx, ConvGeom g, int M,
-                            int RSK, int nk, int fast_a) {
+                            bf16_t* __restrict__ dx, float* __restrict__ ws,
+                            ConvGeom g, int M, int RSK, int nk, int fast_a,
+                            int ksplit) {
   __shared__ bf16_t lds[LDS_ELEMS];
   const int m0 = blockIdx.x * BM;
   const int c0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
+  const int steps = (nk + ksplit - 1) / ksplit;
+  const int kt0 = blockIdx.z * steps;
+  const int kt1 = min(kt0 + steps, nk);
 
   const int arow = t >> 1, ahalf = t & 1;
   int m = m0 + arow;
@@ -208,11 +265,11 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
   int4 areg0, areg1;
-  bf16_t breg[8];
+  __align__(16) bf16_t breg[8];
 
   auto stage_to_regs = [&](int kt) {
     const int k0 = kt * BK;
-    if (fast_a) {  // K % 32 == 0: chunk inside one (r,s)
+    if (fast_a) {  // K % 32 == 0: a 16-chunk stays inside one (r,s)
       int k = k0 + ahalf * 16;
       int rs = k / g.K, kc0 = k - rs * g.K;
       int r = rs / g.S, s = rs - r * g.S;
@@ -231,7 +288,7 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
         areg1 = int4{0, 0, 0, 0};
       }
     } else {
-      bf16_t tmp[16];
+      __align__(16) bf16_t tmp[16];
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         int k = k0 + ahalf * 16 + j;
@@ -252,9 +309,7 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       areg0 = *(int4*)&tmp[0];
       areg1 = *(int4*)&tmp[8];
     }
-    // B: Bs[c][kk] = w[(rs*C + c0+c)*K + kc]
-    // thread: c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0 so
-    // an 8-chunk never crosses an (r,s) boundary)
+    // B: thread: c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0)
     const int bc = t >> 2, bkk = (t & 3) * 8;
     {
       int k = k0 + bkk;
@@ -276,15 +331,17 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     *(int4*)&Bs[bc * LP + bkk] = *(int4*)breg;
   };
 
-  stage_to_regs(0);
-  regs_to_lds(0);
+  if (kt0 < kt1) {
+    stage_to_regs(kt0);
+    regs_to_lds(0);
+  }
   __syncthreads();
-  for (int kt = 0; kt < nk; ++kt) {
-    int cur = kt & 1;
-    if (kt + 1 < nk) stage_to_regs(kt + 1);
+  for (int kt = kt0; kt < kt1; ++kt) {
+    int cur = (kt - kt0) & 1;
+    if (kt + 1 < kt1) stage_to_regs(kt + 1);
     mfma_tile(lds, AS_OFF(cur), BS_OFF(cur), wr, wc, acc);
     __syncthreads();
-    if (kt + 1 < nk) {
+    if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
       __syncthreads();
     }
@@ -299,37 +356,42 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + mr * 16 + fi.quad * 4 + r;
         int col = c0 + wc * 32 + nr * 16 + fi.half;
-        if (row < M && col < g.C)
-          dx[(long)row * g.C + col] = f2bf(acc[mr][nr][r]);
+        if (row < M && col < g.C) {
+          if (ksplit > 1)
+            ws[((long)blockIdx.z * M + row) * g.C + col] = acc[mr][nr][r];
+          else
+            dx[(long)row * g.C + col] = f2bf(acc[mr][nr][r]);
+        }
       }
 }
 
 // ============================== backward weight ==============================
-// dW[(r,s,c), k] += sum_m X[m -> (n,hi,wi,c)] * dY[m, k], split over m-slices
-// with fp32 atomics (dW pre-zeroed). Tile 64(CRS) x 64(K) x 32(m), waves 2x2.
+// dW[(r,s,c), k] = sum_m X[m -> (n,hi,wi,c)] * dY[m, k]; the m-reduction is
+// split over blockIdx.z into fp32 partial slabs (reduced by reduce_slabs_f32).
+// Tile 64(CRS) x 64(K) x 64(m) — BK=64 halves barrier count vs the conv tiles.
 
 #define WBM 64
-#define WLDS_ELEMS (2 * WBM * LP + 2 * BN * LP)
-#define WAS_OFF(buf) ((buf) * WBM * LP)
-#define WBS_OFF(buf) (2 * WBM * LP + (buf) * BN * LP)
+#define WBK 64
+#define WLP (WBK + 8)
+#define WAS_OFF(buf) ((buf) * WBM * WLP)
+#define WBS_OFF(buf) (2 * WBM * WLP + (buf) * BN * WLP)
+#define WLDS_ELEMS (2 * WBM * WLP + 2 * BN * WLP)
 
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
                               const bf16_t* __restrict__ x,
-                              float* __restrict__ dw, ConvGeom g, int M,
+                              float* __restrict__ ws, ConvGeom g, int M,
                               int CRS, int slice_len, int fast_a) {
   __shared__ bf16_t lds[WLDS_ELEMS];
-  const int rs0 = blockIdx.x * WBM;   // CRS rows
-  const int ko0 = blockIdx.y * BN;    // Kout cols
+  const int rs0 = blockIdx.x * WBM;
+  const int ko0 = blockIdx.y * BN;
   const int ms = blockIdx.z * slice_len;
   const int me = min(ms + slice_len, M);
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
 
-  // staging mapping: mm = t>>3 (0..31 reduction cols), grp = t&7 (8-row group)
+  // staging: mm2 = t>>3 (0..31, two cols: mm2 and mm2+32), grp = t&7
   const int amm = t >> 3, agrp = t & 7;
-  // A rows rs0 + agrp*8 + j (j=0..7): when C % 8 == 0 the 8-row group stays in
-  // ONE (r,s) with contiguous c -> one b128 gather per m (fast path)
   int r_ = 0, s_ = 0, cbase_ = 0;
   const int rowb = rs0 + agrp * 8;
   bool agrp_ok = rowb < CRS;
@@ -346,12 +408,11 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  __align__(16) bf16_t areg[8];
-  __align__(16) bf16_t breg[8];
+  __align__(16) bf16_t areg[2][8];
+  __align__(16) bf16_t breg[2][8];
   const int HoWo = g.Ho * g.Wo;
 
-  auto stage_to_regs = [&](int m0) {
-    const int m = m0 + amm;
+  auto gather_one = [&](int m, bf16_t* adst, bf16_t* bdst) {
     int n = 0, ho = 0, wo = 0;
     const bool m_ok = m < me;
     if (m_ok) {
@@ -360,16 +421,14 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
       ho = rem / g.Wo;
       wo = rem - ho * g.Wo;
     }
-    // ---- A: 8 rows (rsc) x 1 col (m): b128 gather from x
     if (fast_a) {
       int hi = ho * g.stride - g.pad + r_;
       int wi = wo * g.stride - g.pad + s_;
-      if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
-        *(int4*)areg = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
+      if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        *(int4*)adst = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
                                         * g.C + cbase_];
-      } else {
-        *(int4*)areg = int4{0, 0, 0, 0};
-      }
+      else
+        *(int4*)adst = int4{0, 0, 0, 0};
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -383,55 +442,63 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
           if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
             v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c];
         }
-        areg[j] = v;
+        adst[j] = v;
       }
     }
-    // ---- B: 8 kouts x 1 col (m): b128 from dy (K % 8 == 0)
     if (m_ok && ko0 + agrp * 8 + 8 <= g.K) {
-      *(int4*)breg = *(const int4*)&dy[(long)m * g.K + ko0 + agrp * 8];
+      *(int4*)bdst = *(const int4*)&dy[(long)m * g.K + ko0 + agrp * 8];
     } else if (m_ok) {
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        breg[j] = (ko0 + agrp * 8 + j < g.K)
+        bdst[j] = (ko0 + agrp * 8 + j < g.K)
                       ? dy[(long)m * g.K + ko0 + agrp * 8 + j] : 0;
     } else {
-      *(int4*)breg = int4{0, 0, 0, 0};
+      *(int4*)bdst = int4{0, 0, 0, 0};
     }
+  };
+
+  auto stage_to_regs = [&](int m0) {
+    gather_one(m0 + amm, areg[0], breg[0]);
+    gather_one(m0 + amm + 32, areg[1], breg[1]);
   };
 
   auto regs_to_lds = [&](int buf) {
     bf16_t* As = &lds[WAS_OFF(buf)];
     bf16_t* Bs = &lds[WBS_OFF(buf)];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) As[(agrp * 8 + j) * LP + amm] = areg[j];
+    for (int h = 0; h < 2; ++h)
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Bs[(agrp * 8 + j) * LP + amm] = breg[j];
+      for (int j = 0; j < 8; ++j) {
+        As[(agrp * 8 + j) * WLP + amm + h * 32] = areg[h][j];
+        Bs[(agrp * 8 + j) * WLP + amm + h * 32] = breg[h][j];
+      }
   };
 
-  const int nk = cdiv_i(me - ms, BK);
+  const int nk = cdiv_i(me - ms, WBK);
   stage_to_regs(ms);
   regs_to_lds(0);
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     int cur = kt & 1;
-    if (kt + 1 < nk) stage_to_regs(ms + (kt + 1) * BK);
-    // wave tile 32x32: M_rep=2, N_rep=2
+    if (kt + 1 < nk) stage_to_regs(ms + (kt + 1) * WBK);
     {
       FragIdx fi = frag_idx();
-      const int kb = fi.quad * 8;
       const bf16_t* As = &lds[WAS_OFF(cur)];
       const bf16_t* Bs = &lds[WBS_OFF(cur)];
 #pragma unroll
-      for (int mr = 0; mr < 2; ++mr) {
-        int row = wr * 32 + mr * 16 + fi.half;
-        bf16x8 a = *(const bf16x8*)&As[row * LP + kb];
+      for (int kh = 0; kh < 2; ++kh) {
+        const int kb = fi.quad * 8 + kh * 32;
 #pragma unroll
-        for (int nr = 0; nr < 2; ++nr) {
-          int col = wc * 32 + nr * 16 + fi.half;
-          bf16x8 b = *(const bf16x8*)&Bs[col * LP + kb];
-          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
-                                                                acc[mr][nr],
-                                                                0, 0, 0);
+        for (int mr = 0; mr < 2; ++mr) {
+          int row = wr * 32 + mr * 16 + fi.half;
+          bf16x8 a = *(const bf16x8*)&As[row * WLP + kb];
+#pragma unroll
+          for (int nr = 0; nr < 2; ++nr) {
+            int col = wc * 32 + nr * 16 + fi.half;
+            bf16x8 b = *(const bf16x8*)&Bs[col * WLP + kb];
+            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[mr][nr], 0, 0, 0);
+          }
         }
       }
     }
@@ -443,6 +510,7 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   }
 
   FragIdx fi = frag_idx();
+  const long slab = (long)blockIdx.z * CRS * g.K;
 #pragma unroll
   for (int mr = 0; mr < 2; ++mr)
 #pragma unroll
@@ -452,58 +520,112 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
         int row = rs0 + wr * 32 + mr * 16 + fi.quad * 4 + r;
         int col = ko0 + wc * 32 + nr * 16 + fi.half;
         if (row < CRS && col < g.K)
-          atomicAdd(&dw[(long)row * g.K + col], acc[mr][nr][r]);
+          ws[slab + (long)row * g.K + col] = acc[mr][nr][r];
       }
 }
 
 // ============================== launchers ==============================
 
+static int pick_ksplit(int nblocks, int nk) {
+  // aim for >= 512 workgroups (2 per CU) without shredding the K loop
+  int ks = 1;
+  while (ks < 8 && nblocks * ks < 512 && nk / (ks * 2) >= 4) ks *= 2;
+  return ks;
+}
+
 extern "C" {
 
-void cilfw_conv2d_fwd(const void* x, const void* w, void* y,
+void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
                       int N, int H, int W, int C, int K, int R, int S,
-                      int stride, int pad, int Ho, int Wo, void* stream) {
+                      int stride, int pad, int Ho, int Wo, int ksplit,
+                      void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int nk = cdiv(CRS, BK);
   int fast_a = (C % BK == 0);
-  dim3 grid(cdiv(M, BM), cdiv(K, BN));
+  dim3 grid(cdiv(M, BM), cdiv(K, BN), ksplit);
   hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(NTHREADS), 0,
                      (hipStream_t)stream, (const bf16_t*)x, (const bf16_t*)w,
-                     (bf16_t*)y, g, M, CRS, nk, fast_a);
+                     (bf16_t*)y, (float*)ws, g, M, CRS, nk, fast_a, ksplit);
+  if (ksplit > 1) {
+    long len = (long)M * K;
+    hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                       dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                       dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                       (bf16_t*)y, ksplit, len);
+  }
 }
 
-void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx,
+int cilfw_conv2d_fwd_ksplit(int N, int C, int K, int R, int S, int Ho,
+                            int Wo) {
+  int M = N * Ho * Wo;
+  int nk = cdiv(C * R * S, BK);
+  return pick_ksplit(cdiv(M, BM) * cdiv(K, BN), nk);
+}
+
+void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
                            int N, int H, int W, int C, int K, int R, int S,
-                           int stride, int pad, int Ho, int Wo, void* stream) {
+                           int stride, int pad, int Ho, int Wo, int ksplit,
+                           void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * H * W;
   int RSK = R * S * K;
   int nk = cdiv(RSK, BK);
   int fast_a = (K % BK == 0);
-  dim3 grid(cdiv(M, BM), cdiv(C, BN));
+  dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
   hipLaunchKernelGGL(conv2d_bwd_data_kernel, grid, dim3(NTHREADS), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)w,
-                     (bf16_t*)dx, g, M, RSK, nk, fast_a);
+                     (bf16_t*)dx, (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  if (ksplit > 1) {
+    long len = (long)M * C;
+    hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                       dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                       dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                       (bf16_t*)dx, ksplit, len);
+  }
+}
+
+int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
+                                 int S) {
+  int M = N * H * W;
+  int nk = cdiv(R * S * K, BK);
+  return pick_ksplit(cdiv(M, BM) * cdiv(C, BN), nk);
 }
 
 void cilfw_conv2d_bwd_weight(const void* dy, const void* x, void* dw,
-                             int N, int H, int W, int C, int K, int R, int S,
-                             int stride, int pad, int Ho, int Wo,
-                             void* stream) {
+                             void* ws, int N, int H, int W, int C, int K,
+                             int R, int S, int stride, int pad, int Ho,
+                             int Wo, int nslices, void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
-  (void)hipMemsetAsync(dw, 0, (size_t)CRS * K * sizeof(float),
-                       (hipStream_t)stream);
-  int slice_len = 4096;
-  int nslices = cdiv(M, slice_len);
+  int slice_len = cdiv(M, nslices);
+  slice_len = cdiv(slice_len, WBK) * WBK;
   int fast_a = (C % 8 == 0);
   dim3 grid(cdiv(CRS, WBM), cdiv(K, BN), nslices);
   hipLaunchKernelGGL(conv2d_bwd_weight_kernel, grid, dim3(NTHREADS), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)x,
-                     (float*)dw, g, M, CRS, slice_len, fast_a);
+                     (float*)ws, g, M, CRS, slice_len, fast_a);
+  long len = (long)CRS * K;
+  hipLaunchKernelGGL(reduce_slabs_f32_kernel,
+                     dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                     dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                     (float*)dw, nslices, len);
+}
+
+int cilfw_conv2d_bwd_weight_nslices(int N, int C, int K, int R, int S,
+                                    int Ho, int Wo) {
+  int M = N * Ho * Wo;
+  int CRS = C * R * S;
+  int nblocks = cdiv(CRS, WBM) * cdiv(K, BN);
+  int target = 512;
+  int ns = cdiv(target, max(nblocks, 1));
+  int max_ns = max(cdiv(M, WBK * 8), 1);  // keep >= 8 K-steps per slice
+  if (ns > max_ns) ns = max_ns;
+  if (ns < 1) ns = 1;
+  if (ns > 64) ns = 64;
+  return ns;
 }
 
 }  // extern "C"
