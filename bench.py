@@ -65,6 +65,8 @@ def main():
         teacher.fc.heads = nn.ModuleList(list(teacher.fc.heads)[:9])  # 90 classes
         teacher.freeze(["all"])
         teacher = teacher.to(device)
+        if dtype == torch.bfloat16:
+            teacher.cast_compute_weights_(dtype)
 
     engine = DataParallelEngine(model, bucket_mb=25.0)
     opt = FlatSGD(engine, lr=0.1, momentum=0.9, weight_decay=5e-4)

@@ -170,6 +170,8 @@ def run(args):
         if start_task < len(scenario_train):
             teacher = model.copy().to(device)
             teacher.freeze(["all"])
+            if compute_dtype(args) == torch.bfloat16:
+                teacher.cast_compute_weights_(torch.bfloat16)
         print(f"resumed from {args.resume}: start_task={start_task}, "
               f"known={args.known_classes}, acc1s={acc1s}")
 
@@ -208,6 +210,8 @@ def run(args):
 
         teacher = model.copy()
         teacher.freeze(["all"])
+        if compute_dtype(args) == torch.bfloat16:
+            teacher.cast_compute_weights_(torch.bfloat16)
 
         features = extract_task_features(model, dataset_train, device, args)
         rx, ry, rt = dataset_train.get_raw_samples()

@@ -115,6 +115,21 @@ class CilModel(nn.Module):
             raise AttributeError(f"unknown submodules to freeze: {missing}")
         return freezed
 
+    @torch.no_grad()
+    def cast_compute_weights_(self, dtype):
+        """Convert conv/linear weights to the compute dtype in place — used on
+        the FROZEN teacher so its forward skips the per-step fp32->bf16 casts
+        (the fp32 masters only matter for models that train)."""
+        from .layers import Conv2d
+        for m in self.modules():
+            if isinstance(m, Conv2d):
+                m.weight.data = m.weight.data.to(dtype)
+        if self.fc is not None:
+            for h in self.fc.heads:
+                h.weight.data = h.weight.data.to(dtype)
+                h.bias.data = h.bias.data.to(dtype)
+        return self
+
     def prev_model_adaption(self, nb_classes):
         if self.fc is None:
             self.fc = CilClassifier(self.feature_dim, nb_classes)
