@@ -37,6 +37,9 @@ def main():
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--no_kd", action="store_true",
                    help="drop the teacher/KD part of the step")
+    p.add_argument("--no_graph", dest="graph", action="store_false",
+                   default=True,
+                   help="disable hipGraph capture of the training step")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -92,13 +95,38 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph capture: the whole step (fwd + teacher fwd + losses + backward
+    # incl. RCCL all-reduce + fused SGD) replays as ONE graph launch —
+    # MI355X-idiomatic replacement for a tracing compiler.
+    run_step = step
+    graphed = False
+    if args.graph and use_cuda:
+        try:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                step()
+            torch.cuda.current_stream().wait_stream(s)
+            gr = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gr):
+                step()
+            run_step = gr.replay
+            graphed = True
+        except Exception as e:  # eager fallback, report it
+            print(f"[bench] graph capture failed ({type(e).__name__}: {e}); "
+                  f"running eager", flush=True)
+            run_step = step
+
+    for _ in range(3):
+        run_step()
+
     if world > 1:
         dist.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        run_step()
     if use_cuda:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -126,7 +154,7 @@ def main():
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": B * world,
                        "input_size": args.input_size,
-                       "kd": not args.no_kd,
+                       "kd": not args.no_kd, "hip_graph": graphed,
                        "parallelism": f"dp{world}"},
         }))
     if world > 1:

@@ -38,7 +38,7 @@ _PROTOS = {
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_bn_fwd": [c_vp] * 9 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
-    "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
+    "cilfw_bn_bwd": [c_vp] * 8 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_downsample_a_fwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
@@ -54,7 +54,7 @@ _PROTOS = {
     "cilfw_ce_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_fwd": [c_vp] * 5 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
-    "cilfw_sgd_step": [c_vp] * 3 + [c_l, c_f, c_f, c_f, c_vp],
+    "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_f, c_f, c_f, c_vp],
     "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_herding_select": [c_vp] * 3 + [c_i] * 3 + [c_vp],
 }
@@ -181,15 +181,14 @@ def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training):
     C = x.shape[-1]
     M = x.numel() // C
     dx = torch.empty_like(x)
-    dgamma = torch.empty(C, dtype=torch.float32, device=x.device)
-    dbeta = torch.empty(C, dtype=torch.float32, device=x.device)
+    dgb = torch.empty(2 * C, dtype=torch.float32, device=x.device)
     gf = gamma.float().contiguous()
     _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(gf),
-                      _ptr(mean), _ptr(invstd), _ptr(dgamma), _ptr(dbeta),
+                      _ptr(mean), _ptr(invstd), _ptr(dgb),
                       c_l(M), c_i(C), c_i(1 if relu else 0),
                       c_i(1 if training else 0), _stream())
     _check("bn_bwd")
-    return dx, dgamma, dbeta
+    return dx, dgb[:C], dgb[C:]
 
 
 # ------------------------------------------------------------------ elementwise
@@ -353,9 +352,10 @@ def kd_bwd(ps, pt, T, dloss):
 
 # ------------------------------------------------------------- optimizer / misc
 
-def sgd_step(p, g, m, lr, momentum, wd):
-    _lib.cilfw_sgd_step(_ptr(p), _ptr(g), _ptr(m), c_l(p.numel()), c_f(lr),
-                        c_f(momentum), c_f(wd), _stream())
+def sgd_step(p, g, m, lr, momentum, wd, p_bf16=None):
+    _lib.cilfw_sgd_step(_ptr(p), _ptr(g), _ptr(m), _ptr(p_bf16),
+                        c_l(p.numel()), c_f(lr), c_f(momentum), c_f(wd),
+                        _stream())
     _check("sgd_step")
 
 

@@ -138,7 +138,8 @@ void kd_bwd_kernel(const float* __restrict__ ps, const float* __restrict__ pt,
 
 __global__ __launch_bounds__(NT)
 void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
-                float* __restrict__ m, long n, float lr, float mu, float wd) {
+                float* __restrict__ m, bf16_t* __restrict__ pb, long n,
+                float lr, float mu, float wd) {
   long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 4;
   if (i0 >= n) return;
   if (i0 + 4 <= n) {
@@ -156,11 +157,16 @@ void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
     }
     *(float4*)&p[i0] = pv;
     *(float4*)&m[i0] = mv;
+    if (pb != nullptr) {  // keep the bf16 compute mirror in sync in-kernel
+      bf16_t o[4] = {f2bf(pe[0]), f2bf(pe[1]), f2bf(pe[2]), f2bf(pe[3])};
+      *(int2*)&pb[i0] = *(int2*)o;
+    }
   } else {
     for (long i = i0; i < n; ++i) {
       float gg = g[i] + wd * p[i];
       m[i] = mu * m[i] + gg;
       p[i] -= lr * m[i];
+      if (pb != nullptr) pb[i] = f2bf(p[i]);
     }
   }
 }
@@ -318,12 +324,12 @@ void cilfw_kd_bwd(const void* ps, const void* pt, const void* dloss, void* ds,
                      M, T);
 }
 
-void cilfw_sgd_step(void* p, const void* g, void* m, long n, float lr,
-                    float mu, float wd, void* stream) {
+void cilfw_sgd_step(void* p, const void* g, void* m, void* pb, long n,
+                    float lr, float mu, float wd, void* stream) {
   long blocks = cdiv((long)n, (long)NT * 4);
   hipLaunchKernelGGL(sgd_kernel, dim3((int)blocks), dim3(NT), 0,
                      (hipStream_t)stream, (float*)p, (const float*)g,
-                     (float*)m, n, lr, mu, wd);
+                     (float*)m, (bf16_t*)pb, n, lr, mu, wd);
 }
 
 void cilfw_topk_correct(const void* logits, const void* targets, void* counts,

@@ -450,17 +450,19 @@ void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
 
 void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                   const void* gamma, const void* mean, const void* invstd,
-                  void* dgamma, void* dbeta, long M, int C, int relu,
+                  void* dgb, long M, int C, int relu,
                   int training, void* stream) {
+  // dgb: ONE fp32 buffer [dgamma | dbeta] (single memset, single allocation)
   hipStream_t st = (hipStream_t)stream;
-  (void)hipMemsetAsync(dgamma, 0, C * sizeof(float), st);
-  (void)hipMemsetAsync(dbeta, 0, C * sizeof(float), st);
+  float* dgamma = (float*)dgb;
+  float* dbeta = dgamma + C;
+  (void)hipMemsetAsync(dgb, 0, 2 * C * sizeof(float), st);
   int rows_per_blk = 256;
   dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
   hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
                      (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
-                     (const float*)mean, (const float*)invstd, (float*)dgamma,
-                     (float*)dbeta, M, C, rows_per_blk, relu);
+                     (const float*)mean, (const float*)invstd, dgamma,
+                     dbeta, M, C, rows_per_blk, relu);
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),

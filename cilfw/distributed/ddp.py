@@ -85,6 +85,15 @@ class DataParallelEngine:
         if _dist_active():
             dist.broadcast(self.flat_params, src=0, group=self.group)
 
+        # bf16 compute mirror: conv/linear forwards read these views instead of
+        # casting the fp32 masters each step; the fused SGD kernel keeps the
+        # mirror in sync in the same launch (cilfw/csrc/loss.hip sgd_kernel)
+        self.flat_bf16 = None
+        if self.flat_params.is_cuda:
+            self.flat_bf16 = self.flat_params.to(torch.bfloat16)
+            for p, off in zip(self.params, self._offsets):
+                p._cilfw_bf16 = self.flat_bf16[off:off + p.numel()].view_as(p)
+
     def _make_hook(self, i):
         bi = self._param_bucket[i]
 
@@ -142,3 +151,5 @@ class DataParallelEngine:
         for p in self.params:
             p.data = p.data.clone()
             p.grad = None
+            if hasattr(p, "_cilfw_bf16"):
+                del p._cilfw_bf16

@@ -38,7 +38,11 @@ class Conv2dNHWC(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, stride, padding):
         ctx.stride, ctx.padding = stride, padding
-        wc = w.to(x.dtype)
+        # engine-maintained bf16 mirror (updated inside the fused SGD kernel)
+        # avoids a per-step cast of the fp32 master
+        wc = getattr(w, "_cilfw_bf16", None)
+        if wc is None or wc.dtype != x.dtype:
+            wc = w.to(x.dtype)
         ctx.save_for_backward(x, wc)
         ctx.w_dtype = w.dtype
         if use_hip(x):
@@ -246,7 +250,9 @@ class LinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b):
-        wc = w.to(x.dtype)
+        wc = getattr(w, "_cilfw_bf16", None)
+        if wc is None or wc.dtype != x.dtype:
+            wc = w.to(x.dtype)
         ctx.save_for_backward(x, wc)
         ctx.w_dtype = w.dtype
         ctx.has_bias = b is not None

@@ -28,14 +28,18 @@ class FlatSGD:
     def step(self):
         p, g, m = (self.engine.flat_params, self.engine.flat_grads,
                    self.momentum_buf)
+        mirror = getattr(self.engine, "flat_bf16", None)
         if use_hip(p):
-            ext().sgd_step(p, g, m, self.lr, self.momentum, self.weight_decay)
+            ext().sgd_step(p, g, m, self.lr, self.momentum, self.weight_decay,
+                           mirror)
             return
-        # CPU reference: same math, fused at the ATen level
+        # CPU reference (and CILFW_FORCE_TORCH): same math, fused ATen ops
         if self.weight_decay != 0:
             g = g.add(p, alpha=self.weight_decay)
         m.mul_(self.momentum).add_(g)
         p.add_(m, alpha=-self.lr)
+        if mirror is not None:
+            mirror.copy_(p)
 
     def zero_grad(self):
         self.engine.zero_grad()
