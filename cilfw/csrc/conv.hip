@@ -4,9 +4,11 @@
 // Formulation: Y[M=N*Ho*Wo, K] = im2col(X)[M, C*R*S] @ W[C*R*S, K], W stored
 // (R,S,C,K) so the B operand is dense with k-index (r*S+s)*C+c.
 //
-// Tile: 128(M) x 64(Kout) x 32(K-step), 4 waves as 2x2, each wave 64x32 via
+// Tile: 128(M) x 64(Kout) x BKT(K-step), 4 waves as 2x2, each wave 64x32 via
 // mfma_f32_16x16x32_bf16 (M_rep=4, N_rep=2), fp32 accumulators, double-buffered
-// LDS with +16B row padding (bank-conflict-free ds_read_b128).
+// LDS with +16B row padding (bank-conflict-free ds_read_b128). BKT is a
+// template parameter: 64 when the K loop is long enough (halves the barrier
+// count, 16 MFMAs between barriers), 32 for short-K convs (stems, 1x1).
 //
 // Late layers have small M (batch x 4x4 spatial) and huge K-dim (C*R*S up to
 // 4608): too few workgroups to fill 256 CUs. When (M-blocks x K-blocks) is
@@ -18,36 +20,11 @@
 
 #define BM 128
 #define BN 64
-#define BK 32
-#define LP (BK + 8)          // LDS row pitch in bf16 elements (+16B pad)
 #define NTHREADS 256
-
-#define AS_OFF(buf) ((buf) * BM * LP)
-#define BS_OFF(buf) (2 * BM * LP + (buf) * BN * LP)
-#define LDS_ELEMS (2 * BM * LP + 2 * BN * LP)
 
 struct ConvGeom {
   int N, H, W, C, K, R, S, stride, pad, Ho, Wo;
 };
-
-// ---- shared MFMA core: given staged As/Bs, accumulate 4x2 fragments ----
-DEV void mfma_tile(const bf16_t* lds, int a_off, int b_off, int wr, int wc,
-                   f32x4 acc[4][2]) {
-  FragIdx fi = frag_idx();
-  const int kb = fi.quad * 8;
-#pragma unroll
-  for (int mr = 0; mr < 4; ++mr) {
-    int row = wr * 64 + mr * 16 + fi.half;
-    bf16x8 a = *(const bf16x8*)&lds[a_off + row * LP + kb];
-#pragma unroll
-    for (int nr = 0; nr < 2; ++nr) {
-      int col = wc * 32 + nr * 16 + fi.half;
-      bf16x8 b = *(const bf16x8*)&lds[b_off + col * LP + kb];
-      acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mr][nr],
-                                                            0, 0, 0);
-    }
-  }
-}
 
 // sum ksplit fp32 slabs [ns][len] -> bf16 out[len]
 __global__ __launch_bounds__(NTHREADS)
@@ -96,18 +73,21 @@ void reduce_slabs_f32_kernel(const float* __restrict__ ws,
 
 // ============================== forward ==============================
 
+template <int BKT>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        const bf16_t* __restrict__ w,
                        bf16_t* __restrict__ y, float* __restrict__ ws,
                        ConvGeom g, int M, int CRS, int nk, int fast_a,
                        int ksplit) {
-  __shared__ bf16_t lds[LDS_ELEMS];
+  constexpr int NQ = BKT / 32;      // 16-elem chunks per thread in A staging
+  constexpr int LPX = BKT + 8;
+  __shared__ bf16_t lds[2 * BM * LPX + 2 * BN * LPX];
+  const int AS0 = 0, BS0 = 2 * BM * LPX;
   const int m0 = blockIdx.x * BM;
   const int ko0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
-  // this z-slice's K-step range
   const int steps = (nk + ksplit - 1) / ksplit;
   const int kt0 = blockIdx.z * steps;
   const int kt1 = min(kt0 + steps, nk);
@@ -131,68 +111,75 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  int4 areg0, areg1;
-  __align__(16) bf16_t breg[8];
+  int4 areg[2 * NQ];
+  __align__(16) bf16_t breg[NQ][8];
 
   auto stage_to_regs = [&](int kt) {
-    const int k0 = kt * BK;
-    if (fast_a) {
-      int k = k0 + ahalf * 16;
-      int rs = k / g.C, c0 = k - rs * g.C;
-      int r = rs / g.S, s = rs - r * g.S;
-      int hi = ahb + r, wi = awb + s;
-      if (arow_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
-        const int4* src = (const int4*)&x[(((long)an * g.H + hi) * g.W + wi)
-                                          * g.C + c0];
-        areg0 = src[0];
-        areg1 = src[1];
-      } else {
-        areg0 = int4{0, 0, 0, 0};
-        areg1 = int4{0, 0, 0, 0};
-      }
-    } else {
-      __align__(16) bf16_t tmp[16];
+    const int k0 = kt * BKT;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        int k = k0 + ahalf * 16 + j;
-        bf16_t v = 0;
-        if (arow_ok && k < CRS) {
-          int rs = k / g.C, c = k - rs * g.C;
-          int r = rs / g.S, s = rs - r * g.S;
-          int hi = ahb + r, wi = awb + s;
-          if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-            v = x[(((long)an * g.H + hi) * g.W + wi) * g.C + c];
+    for (int q = 0; q < NQ; ++q) {
+      // A chunk: 16 elems at col q*32 + ahalf*16 (within one (r,s): C%32==0)
+      if (fast_a) {
+        int k = k0 + q * 32 + ahalf * 16;
+        int rs = k / g.C, c0 = k - rs * g.C;
+        int r = rs / g.S, s = rs - r * g.S;
+        int hi = ahb + r, wi = awb + s;
+        if (arow_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
+          const int4* src = (const int4*)&x[(((long)an * g.H + hi) * g.W + wi)
+                                            * g.C + c0];
+          areg[2 * q] = src[0];
+          areg[2 * q + 1] = src[1];
+        } else {
+          areg[2 * q] = int4{0, 0, 0, 0};
+          areg[2 * q + 1] = int4{0, 0, 0, 0};
         }
-        tmp[j] = v;
+      } else {
+        __align__(16) bf16_t tmp[16];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int k = k0 + q * 32 + ahalf * 16 + j;
+          bf16_t v = 0;
+          if (arow_ok && k < CRS) {
+            int rs = k / g.C, c = k - rs * g.C;
+            int r = rs / g.S, s = rs - r * g.S;
+            int hi = ahb + r, wi = awb + s;
+            if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+              v = x[(((long)an * g.H + hi) * g.W + wi) * g.C + c];
+          }
+          tmp[j] = v;
+        }
+        areg[2 * q] = *(int4*)&tmp[0];
+        areg[2 * q + 1] = *(int4*)&tmp[8];
       }
-      areg0 = *(int4*)&tmp[0];
-      areg1 = *(int4*)&tmp[8];
-    }
-    // B: thread: kk = t>>3 (0..31), ng = t&7 (8 couts; K % 8 == 0)
-    const int bkk = t >> 3, bng = t & 7;
-    {
+      // B chunk: thread kk = t>>3 (+32q), ng = t&7 (8 couts; K % 8 == 0)
+      const int bkk = (t >> 3) + q * 32;
+      const int bng = t & 7;
       int k = k0 + bkk;
       if (k < CRS && ko0 + bng * 8 + 8 <= g.K) {
-        *(int4*)breg = *(const int4*)&w[(long)k * g.K + ko0 + bng * 8];
+        *(int4*)breg[q] = *(const int4*)&w[(long)k * g.K + ko0 + bng * 8];
       } else if (k < CRS && ko0 + bng * 8 < g.K) {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          breg[j] = (ko0 + bng * 8 + j < g.K)
-                        ? w[(long)k * g.K + ko0 + bng * 8 + j] : 0;
+          breg[q][j] = (ko0 + bng * 8 + j < g.K)
+                           ? w[(long)k * g.K + ko0 + bng * 8 + j] : 0;
       } else {
-        *(int4*)breg = int4{0, 0, 0, 0};
+        *(int4*)breg[q] = int4{0, 0, 0, 0};
       }
     }
   };
 
   auto regs_to_lds = [&](int buf) {
-    bf16_t* As = &lds[AS_OFF(buf)];
-    bf16_t* Bs = &lds[BS_OFF(buf)];
-    *(int4*)&As[arow * LP + ahalf * 16] = areg0;
-    *(int4*)&As[arow * LP + ahalf * 16 + 8] = areg1;
-    const int bkk = t >> 3, bng = t & 7;
+    bf16_t* As = &lds[AS0 + buf * BM * LPX];
+    bf16_t* Bs = &lds[BS0 + buf * BN * LPX];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) Bs[(bng * 8 + j) * LP + bkk] = breg[j];
+    for (int q = 0; q < NQ; ++q) {
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16] = areg[2 * q];
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16 + 8] = areg[2 * q + 1];
+      const int bkk = (t >> 3) + q * 32;
+      const int bng = t & 7;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) Bs[(bng * 8 + j) * LPX + bkk] = breg[q][j];
+    }
   };
 
   if (kt0 < kt1) {
@@ -200,10 +187,30 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     regs_to_lds(0);
   }
   __syncthreads();
+  FragIdx fi = frag_idx();
   for (int kt = kt0; kt < kt1; ++kt) {
     int cur = (kt - kt0) & 1;
     if (kt + 1 < kt1) stage_to_regs(kt + 1);
-    mfma_tile(lds, AS_OFF(cur), BS_OFF(cur), wr, wc, acc);
+    {
+      const bf16_t* As = &lds[AS0 + cur * BM * LPX];
+      const bf16_t* Bs = &lds[BS0 + cur * BN * LPX];
+#pragma unroll
+      for (int q = 0; q < NQ; ++q) {
+        const int kb = fi.quad * 8 + q * 32;
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr) {
+          int row = wr * 64 + mr * 16 + fi.half;
+          bf16x8 a = *(const bf16x8*)&As[row * LPX + kb];
+#pragma unroll
+          for (int nr = 0; nr < 2; ++nr) {
+            int col = wc * 32 + nr * 16 + fi.half;
+            bf16x8 b = *(const bf16x8*)&Bs[col * LPX + kb];
+            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[mr][nr], 0, 0, 0);
+          }
+        }
+      }
+    }
     __syncthreads();
     if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
@@ -211,7 +218,6 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     }
   }
 
-  FragIdx fi = frag_idx();
 #pragma unroll
   for (int mr = 0; mr < 4; ++mr)
 #pragma unroll
@@ -232,13 +238,17 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 // ============================== backward data ==============================
 // dX[M=N*H*W, C] = gather(dY)[M, R*S*K] @ B where B[(r,s,k)][c] = W[r,s,c,k].
 
+template <int BKT>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
                             const bf16_t* __restrict__ w,
                             bf16_t* __restrict__ dx, float* __restrict__ ws,
                             ConvGeom g, int M, int RSK, int nk, int fast_a,
                             int ksplit) {
-  __shared__ bf16_t lds[LDS_ELEMS];
+  constexpr int NQ = BKT / 32;
+  constexpr int LPX = BKT + 8;
+  __shared__ bf16_t lds[2 * BM * LPX + 2 * BN * LPX];
+  const int AS0 = 0, BS0 = 2 * BM * LPX;
   const int m0 = blockIdx.x * BM;
   const int c0 = blockIdx.y * BN;
   const int t = threadIdx.x;
@@ -264,71 +274,76 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  int4 areg0, areg1;
-  __align__(16) bf16_t breg[8];
+  int4 areg[2 * NQ];
+  __align__(16) bf16_t breg[NQ][8];
 
   auto stage_to_regs = [&](int kt) {
-    const int k0 = kt * BK;
-    if (fast_a) {  // K % 32 == 0: a 16-chunk stays inside one (r,s)
-      int k = k0 + ahalf * 16;
-      int rs = k / g.K, kc0 = k - rs * g.K;
-      int r = rs / g.S, s = rs - r * g.S;
-      int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
-      bool ok = arow_ok && ho2 >= 0 && wo2 >= 0 &&
-                (ho2 % g.stride) == 0 && (wo2 % g.stride) == 0;
-      int ho = ho2 / g.stride, wo = wo2 / g.stride;
-      ok = ok && ho < g.Ho && wo < g.Wo;
-      if (ok) {
-        const int4* src = (const int4*)&dy[(((long)an * g.Ho + ho) * g.Wo + wo)
-                                           * g.K + kc0];
-        areg0 = src[0];
-        areg1 = src[1];
-      } else {
-        areg0 = int4{0, 0, 0, 0};
-        areg1 = int4{0, 0, 0, 0};
-      }
-    } else {
-      __align__(16) bf16_t tmp[16];
+    const int k0 = kt * BKT;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        int k = k0 + ahalf * 16 + j;
-        bf16_t v = 0;
-        if (arow_ok && k < RSK) {
-          int rs = k / g.K, kc = k - rs * g.K;
-          int r = rs / g.S, s = rs - r * g.S;
-          int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
-          if (ho2 >= 0 && wo2 >= 0 && (ho2 % g.stride) == 0 &&
-              (wo2 % g.stride) == 0) {
-            int ho = ho2 / g.stride, wo = wo2 / g.stride;
-            if (ho < g.Ho && wo < g.Wo)
-              v = dy[(((long)an * g.Ho + ho) * g.Wo + wo) * g.K + kc];
-          }
+    for (int q = 0; q < NQ; ++q) {
+      if (fast_a) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
+        int k = k0 + q * 32 + ahalf * 16;
+        int rs = k / g.K, kc0 = k - rs * g.K;
+        int r = rs / g.S, s = rs - r * g.S;
+        int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
+        bool ok = arow_ok && ho2 >= 0 && wo2 >= 0 &&
+                  (ho2 % g.stride) == 0 && (wo2 % g.stride) == 0;
+        int ho = ho2 / g.stride, wo = wo2 / g.stride;
+        ok = ok && ho < g.Ho && wo < g.Wo;
+        if (ok) {
+          const int4* src = (const int4*)&dy[(((long)an * g.Ho + ho) * g.Wo
+                                              + wo) * g.K + kc0];
+          areg[2 * q] = src[0];
+          areg[2 * q + 1] = src[1];
+        } else {
+          areg[2 * q] = int4{0, 0, 0, 0};
+          areg[2 * q + 1] = int4{0, 0, 0, 0};
         }
-        tmp[j] = v;
+      } else {
+        __align__(16) bf16_t tmp[16];
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int k = k0 + q * 32 + ahalf * 16 + j;
+          bf16_t v = 0;
+          if (arow_ok && k < RSK) {
+            int rs = k / g.K, kc = k - rs * g.K;
+            int r = rs / g.S, s = rs - r * g.S;
+            int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
+            if (ho2 >= 0 && wo2 >= 0 && (ho2 % g.stride) == 0 &&
+                (wo2 % g.stride) == 0) {
+              int ho = ho2 / g.stride, wo = wo2 / g.stride;
+              if (ho < g.Ho && wo < g.Wo)
+                v = dy[(((long)an * g.Ho + ho) * g.Wo + wo) * g.K + kc];
+            }
+          }
+          tmp[j] = v;
+        }
+        areg[2 * q] = *(int4*)&tmp[0];
+        areg[2 * q + 1] = *(int4*)&tmp[8];
       }
-      areg0 = *(int4*)&tmp[0];
-      areg1 = *(int4*)&tmp[8];
-    }
-    // B: thread: c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0)
-    const int bc = t >> 2, bkk = (t & 3) * 8;
-    {
+      // B: thread c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0)
+      const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
       int k = k0 + bkk;
       if (k < RSK && c0 + bc < g.C) {
         int rs = k / g.K, kc = k - rs * g.K;
-        *(int4*)breg = *(const int4*)&w[((long)rs * g.C + c0 + bc) * g.K + kc];
+        *(int4*)breg[q] = *(const int4*)&w[((long)rs * g.C + c0 + bc) * g.K
+                                           + kc];
       } else {
-        *(int4*)breg = int4{0, 0, 0, 0};
+        *(int4*)breg[q] = int4{0, 0, 0, 0};
       }
     }
   };
 
   auto regs_to_lds = [&](int buf) {
-    bf16_t* As = &lds[AS_OFF(buf)];
-    bf16_t* Bs = &lds[BS_OFF(buf)];
-    *(int4*)&As[arow * LP + ahalf * 16] = areg0;
-    *(int4*)&As[arow * LP + ahalf * 16 + 8] = areg1;
-    const int bc = t >> 2, bkk = (t & 3) * 8;
-    *(int4*)&Bs[bc * LP + bkk] = *(int4*)breg;
+    bf16_t* As = &lds[AS0 + buf * BM * LPX];
+    bf16_t* Bs = &lds[BS0 + buf * BN * LPX];
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16] = areg[2 * q];
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16 + 8] = areg[2 * q + 1];
+      const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
+      *(int4*)&Bs[bc * LPX + bkk] = *(int4*)breg[q];
+    }
   };
 
   if (kt0 < kt1) {
@@ -336,10 +351,30 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     regs_to_lds(0);
   }
   __syncthreads();
+  FragIdx fi = frag_idx();
   for (int kt = kt0; kt < kt1; ++kt) {
     int cur = (kt - kt0) & 1;
     if (kt + 1 < kt1) stage_to_regs(kt + 1);
-    mfma_tile(lds, AS_OFF(cur), BS_OFF(cur), wr, wc, acc);
+    {
+      const bf16_t* As = &lds[AS0 + cur * BM * LPX];
+      const bf16_t* Bs = &lds[BS0 + cur * BN * LPX];
+#pragma unroll
+      for (int q = 0; q < NQ; ++q) {
+        const int kb = fi.quad * 8 + q * 32;
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr) {
+          int row = wr * 64 + mr * 16 + fi.half;
+          bf16x8 a = *(const bf16x8*)&As[row * LPX + kb];
+#pragma unroll
+          for (int nr = 0; nr < 2; ++nr) {
+            int col = wc * 32 + nr * 16 + fi.half;
+            bf16x8 b = *(const bf16x8*)&Bs[col * LPX + kb];
+            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[mr][nr], 0, 0, 0);
+          }
+        }
+      }
+    }
     __syncthreads();
     if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
@@ -347,7 +382,6 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     }
   }
 
-  FragIdx fi = frag_idx();
 #pragma unroll
   for (int mr = 0; mr < 4; ++mr)
 #pragma unroll
@@ -390,7 +424,6 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
 
-  // staging: mm2 = t>>3 (0..31, two cols: mm2 and mm2+32), grp = t&7
   const int amm = t >> 3, agrp = t & 7;
   int r_ = 0, s_ = 0, cbase_ = 0;
   const int rowb = rs0 + agrp * 8;
@@ -475,8 +508,10 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   };
 
   const int nk = cdiv_i(me - ms, WBK);
-  stage_to_regs(ms);
-  regs_to_lds(0);
+  if (nk > 0) {
+    stage_to_regs(ms);
+    regs_to_lds(0);
+  }
   __syncthreads();
   for (int kt = 0; kt < nk; ++kt) {
     int cur = kt & 1;
@@ -542,12 +577,20 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
-  int nk = cdiv(CRS, BK);
-  int fast_a = (C % BK == 0);
+  int fast_a = (C % 16 == 0);
+  int use64 = (CRS >= 512);
+  int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(K, BN), ksplit);
-  hipLaunchKernelGGL(conv2d_fwd_kernel, grid, dim3(NTHREADS), 0,
-                     (hipStream_t)stream, (const bf16_t*)x, (const bf16_t*)w,
-                     (bf16_t*)y, (float*)ws, g, M, CRS, nk, fast_a, ksplit);
+  if (use64)
+    hipLaunchKernelGGL((conv2d_fwd_kernel<64>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)x,
+                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
+                       nk, fast_a, ksplit);
+  else
+    hipLaunchKernelGGL((conv2d_fwd_kernel<32>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)x,
+                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
+                       nk, fast_a, ksplit);
   if (ksplit > 1) {
     long len = (long)M * K;
     hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -560,7 +603,8 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
 int cilfw_conv2d_fwd_ksplit(int N, int C, int K, int R, int S, int Ho,
                             int Wo) {
   int M = N * Ho * Wo;
-  int nk = cdiv(C * R * S, BK);
+  int CRS = C * R * S;
+  int nk = cdiv(CRS, CRS >= 512 ? 64 : 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(K, BN), nk);
 }
 
@@ -571,12 +615,20 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * H * W;
   int RSK = R * S * K;
-  int nk = cdiv(RSK, BK);
-  int fast_a = (K % BK == 0);
+  int fast_a = (K % 16 == 0);
+  int use64 = (RSK >= 512);
+  int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
-  hipLaunchKernelGGL(conv2d_bwd_data_kernel, grid, dim3(NTHREADS), 0,
-                     (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)w,
-                     (bf16_t*)dx, (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  if (use64)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
+                       nk, fast_a, ksplit);
+  else
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
+                       nk, fast_a, ksplit);
   if (ksplit > 1) {
     long len = (long)M * C;
     hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -589,7 +641,8 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
 int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
                                  int S) {
   int M = N * H * W;
-  int nk = cdiv(R * S * K, BK);
+  int RSK = R * S * K;
+  int nk = cdiv(RSK, RSK >= 512 ? 64 : 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(C, BN), nk);
 }
 
