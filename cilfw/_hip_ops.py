@@ -36,7 +36,8 @@ _lib.cilfw_error_string.argtypes = [c_i]
 _PROTOS = {
     "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
-    "cilfw_conv2d_bwd_weight": [c_vp] * 4 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 12 + [c_vp],
+    "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
     "cilfw_bn_fwd": [c_vp] * 9 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
     "cilfw_bn_bwd": [c_vp] * 8 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
@@ -136,19 +137,40 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
     return dx
 
 
+_mtable_cache = {}
+
+
+def _mtable(N, Ho, Wo, stride, device):
+    """Packed im2col pixel table mt[m] = n<<20 | ho*stride<<10 | wo*stride,
+    built once per conv geometry and cached for the process lifetime."""
+    key = (N, Ho, Wo, stride, device)
+    mt = _mtable_cache.get(key)
+    if mt is None:
+        assert N < 4096 and Ho * stride < 1024 and Wo * stride < 1024, \
+            "mtable packing limits exceeded"
+        M = N * Ho * Wo
+        mt = torch.empty(M, dtype=torch.int32, device=device)
+        _lib.cilfw_fill_mtable(_ptr(mt), c_i(M), c_i(Ho * Wo), c_i(Wo),
+                               c_i(stride), _stream())
+        _mtable_cache[key] = mt
+    return mt
+
+
 def conv2d_bwd_weight(dy, x, stride, pad, R, S):
     _bf16(dy, "conv2d_bwd_weight.dy")
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
+    mt = _mtable(N, Ho, Wo, stride, dy.device)
     dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
     ns = _lib.cilfw_conv2d_bwd_weight_nslices(N, C, K, R, S, Ho, Wo)
     ws = torch.empty(ns * R * S * C * K, dtype=torch.float32,
                      device=dy.device)
-    _lib.cilfw_conv2d_bwd_weight(_ptr(dy), _ptr(x), _ptr(dw), _ptr(ws),
-                                 c_i(N), c_i(H), c_i(W_), c_i(C), c_i(K),
-                                 c_i(R), c_i(S), c_i(stride), c_i(pad),
-                                 c_i(Ho), c_i(Wo), c_i(ns), _stream())
+    _lib.cilfw_conv2d_bwd_weight(_ptr(dy), _ptr(x), _ptr(mt), _ptr(dw),
+                                 _ptr(ws), c_i(N), c_i(H), c_i(W_), c_i(C),
+                                 c_i(K), c_i(R), c_i(S), c_i(stride),
+                                 c_i(pad), c_i(Ho), c_i(Wo), c_i(ns),
+                                 _stream())
     _check("conv2d_bwd_weight")
     return dw
 

@@ -151,19 +151,15 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         areg[2 * q] = *(int4*)&tmp[0];
         areg[2 * q + 1] = *(int4*)&tmp[8];
       }
-      // B chunk: thread kk = t>>3 (+32q), ng = t&7 (8 couts; K % 8 == 0)
-      const int bkk = (t >> 3) + q * 32;
-      const int bng = t & 7;
-      int k = k0 + bkk;
-      if (k < CRS && ko0 + bng * 8 + 8 <= g.K) {
-        *(int4*)breg[q] = *(const int4*)&w[(long)k * g.K + ko0 + bng * 8];
-      } else if (k < CRS && ko0 + bng * 8 < g.K) {
+      // B chunk: thread owns one Bs row (cout n = t&63) and 8 k's
+      // (kgrp = t>>6): strided reads are coalesced ACROSS lanes (n contiguous)
+      // and the LDS write is one b128 (conflict-free 8-lane groups).
+      const int bn = t & 63, bk8 = (t >> 6) * 8 + q * 32;
+      const bool n_ok = ko0 + bn < g.K;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          breg[q][j] = (ko0 + bng * 8 + j < g.K)
-                           ? w[(long)k * g.K + ko0 + bng * 8 + j] : 0;
-      } else {
-        *(int4*)breg[q] = int4{0, 0, 0, 0};
+      for (int j = 0; j < 8; ++j) {
+        int k = k0 + bk8 + j;
+        breg[q][j] = (n_ok && k < CRS) ? w[(long)k * g.K + ko0 + bn] : 0;
       }
     }
   };
@@ -175,10 +171,8 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     for (int q = 0; q < NQ; ++q) {
       *(int4*)&As[arow * LPX + q * 32 + ahalf * 16] = areg[2 * q];
       *(int4*)&As[arow * LPX + q * 32 + ahalf * 16 + 8] = areg[2 * q + 1];
-      const int bkk = (t >> 3) + q * 32;
-      const int bng = t & 7;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) Bs[(bng * 8 + j) * LPX + bkk] = breg[q][j];
+      const int bn = t & 63, bk8 = (t >> 6) * 8 + q * 32;
+      *(int4*)&Bs[bn * LPX + bk8] = *(int4*)breg[q];
     }
   };
 
@@ -402,7 +396,12 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
 // ============================== backward weight ==============================
 // dW[(r,s,c), k] = sum_m X[m -> (n,hi,wi,c)] * dY[m, k]; the m-reduction is
 // split over blockIdx.z into fp32 partial slabs (reduced by reduce_slabs_f32).
-// Tile 64(CRS) x 64(K) x 64(m) — BK=64 halves barrier count vs the conv tiles.
+// Tile 64(CRS) x 64(K) x 64(m). Staging: each thread OWNS one LDS row (an rsc
+// row for A / a cout for B) and gathers 8 m-columns per chunk — global reads
+// coalesce ACROSS lanes (c / cout contiguous) and each LDS write is one b128
+// (the row-per-8-lane-group pattern is bank-conflict-free). The per-m pixel
+// decomposition comes from a precomputed packed table mt[m] =
+// n<<20 | (ho*stride)<<10 | (wo*stride)  (built once per conv geometry).
 
 #define WBM 64
 #define WBK 64
@@ -414,8 +413,9 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
                               const bf16_t* __restrict__ x,
+                              const int* __restrict__ mt,
                               float* __restrict__ ws, ConvGeom g, int M,
-                              int CRS, int slice_len, int fast_a) {
+                              int CRS, int slice_len) {
   __shared__ bf16_t lds[WLDS_ELEMS];
   const int rs0 = blockIdx.x * WBM;
   const int ko0 = blockIdx.y * BN;
@@ -424,16 +424,18 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
 
-  const int amm = t >> 3, agrp = t & 7;
-  int r_ = 0, s_ = 0, cbase_ = 0;
-  const int rowb = rs0 + agrp * 8;
-  bool agrp_ok = rowb < CRS;
-  if (agrp_ok) {
-    int rs = rowb / g.C;
-    cbase_ = rowb - rs * g.C;
+  // this thread's A row (rsc) and B row (cout); mgrp = which 8 m's per chunk
+  const int row = t & 63, mgrp = t >> 6;
+  int r_ = 0, s_ = 0, c_ = 0;
+  const bool row_ok = rs0 + row < CRS;
+  if (row_ok) {
+    int k = rs0 + row;
+    int rs = k / g.C;
+    c_ = k - rs * g.C;
     r_ = rs / g.S;
     s_ = rs - r_ * g.S;
   }
+  const bool n_ok = ko0 + row < g.K;
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -443,68 +445,40 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
 
   __align__(16) bf16_t areg[2][8];
   __align__(16) bf16_t breg[2][8];
-  const int HoWo = g.Ho * g.Wo;
-
-  auto gather_one = [&](int m, bf16_t* adst, bf16_t* bdst) {
-    int n = 0, ho = 0, wo = 0;
-    const bool m_ok = m < me;
-    if (m_ok) {
-      n = m / HoWo;
-      int rem = m - n * HoWo;
-      ho = rem / g.Wo;
-      wo = rem - ho * g.Wo;
-    }
-    if (fast_a) {
-      int hi = ho * g.stride - g.pad + r_;
-      int wi = wo * g.stride - g.pad + s_;
-      if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-        *(int4*)adst = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
-                                        * g.C + cbase_];
-      else
-        *(int4*)adst = int4{0, 0, 0, 0};
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int row = rowb + j;
-        bf16_t v = 0;
-        if (row < CRS && m_ok) {
-          int rs = row / g.C, c = row - rs * g.C;
-          int r = rs / g.S, s = rs - r * g.S;
-          int hi = ho * g.stride - g.pad + r;
-          int wi = wo * g.stride - g.pad + s;
-          if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-            v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c];
-        }
-        adst[j] = v;
-      }
-    }
-    if (m_ok && ko0 + agrp * 8 + 8 <= g.K) {
-      *(int4*)bdst = *(const int4*)&dy[(long)m * g.K + ko0 + agrp * 8];
-    } else if (m_ok) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        bdst[j] = (ko0 + agrp * 8 + j < g.K)
-                      ? dy[(long)m * g.K + ko0 + agrp * 8 + j] : 0;
-    } else {
-      *(int4*)bdst = int4{0, 0, 0, 0};
-    }
-  };
 
   auto stage_to_regs = [&](int m0) {
-    gather_one(m0 + amm, areg[0], breg[0]);
-    gather_one(m0 + amm + 32, areg[1], breg[1]);
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int mb = m0 + q * 32 + mgrp * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int m = mb + j;
+        bf16_t av = 0, bv = 0;
+        if (m < me) {
+          const int v = mt[m];           // uniform across the wave: broadcast
+          if (row_ok) {
+            const int n = v >> 20;
+            const int hi = ((v >> 10) & 1023) - g.pad + r_;
+            const int wi = (v & 1023) - g.pad + s_;
+            if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+              av = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c_];
+          }
+          if (n_ok) bv = dy[(long)m * g.K + ko0 + row];
+        }
+        areg[q][j] = av;
+        breg[q][j] = bv;
+      }
+    }
   };
 
   auto regs_to_lds = [&](int buf) {
     bf16_t* As = &lds[WAS_OFF(buf)];
     bf16_t* Bs = &lds[WBS_OFF(buf)];
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        As[(agrp * 8 + j) * WLP + amm + h * 32] = areg[h][j];
-        Bs[(agrp * 8 + j) * WLP + amm + h * 32] = breg[h][j];
-      }
+    for (int q = 0; q < 2; ++q) {
+      *(int4*)&As[row * WLP + q * 32 + mgrp * 8] = *(int4*)areg[q];
+      *(int4*)&Bs[row * WLP + q * 32 + mgrp * 8] = *(int4*)breg[q];
+    }
   };
 
   const int nk = cdiv_i(me - ms, WBK);
@@ -525,8 +499,8 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
         const int kb = fi.quad * 8 + kh * 32;
 #pragma unroll
         for (int mr = 0; mr < 2; ++mr) {
-          int row = wr * 32 + mr * 16 + fi.half;
-          bf16x8 a = *(const bf16x8*)&As[row * WLP + kb];
+          int arow = wr * 32 + mr * 16 + fi.half;
+          bf16x8 a = *(const bf16x8*)&As[arow * WLP + kb];
 #pragma unroll
           for (int nr = 0; nr < 2; ++nr) {
             int col = wc * 32 + nr * 16 + fi.half;
@@ -552,11 +526,24 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     for (int nr = 0; nr < 2; ++nr)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = rs0 + wr * 32 + mr * 16 + fi.quad * 4 + r;
+        int orow = rs0 + wr * 32 + mr * 16 + fi.quad * 4 + r;
         int col = ko0 + wc * 32 + nr * 16 + fi.half;
-        if (row < CRS && col < g.K)
-          ws[slab + (long)row * g.K + col] = acc[mr][nr][r];
+        if (orow < CRS && col < g.K)
+          ws[slab + (long)orow * g.K + col] = acc[mr][nr][r];
       }
+}
+
+// packed im2col pixel table: mt[m] = n<<20 | (ho*stride)<<10 | (wo*stride)
+__global__ __launch_bounds__(NTHREADS)
+void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
+                        int stride) {
+  int m = blockIdx.x * NTHREADS + threadIdx.x;
+  if (m >= M) return;
+  int n = m / HoWo;
+  int rem = m - n * HoWo;
+  int ho = rem / Wo;
+  int wo = rem - ho * Wo;
+  mt[m] = (n << 20) | ((ho * stride) << 10) | (wo * stride);
 }
 
 // ============================== launchers ==============================
@@ -646,25 +633,31 @@ int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
   return pick_ksplit(cdiv(M, BM) * cdiv(C, BN), nk);
 }
 
-void cilfw_conv2d_bwd_weight(const void* dy, const void* x, void* dw,
-                             void* ws, int N, int H, int W, int C, int K,
-                             int R, int S, int stride, int pad, int Ho,
+void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
+                             void* dw, void* ws, int N, int H, int W, int C,
+                             int K, int R, int S, int stride, int pad, int Ho,
                              int Wo, int nslices, void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int slice_len = cdiv(M, nslices);
   slice_len = cdiv(slice_len, WBK) * WBK;
-  int fast_a = (C % 8 == 0);
   dim3 grid(cdiv(CRS, WBM), cdiv(K, BN), nslices);
   hipLaunchKernelGGL(conv2d_bwd_weight_kernel, grid, dim3(NTHREADS), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)x,
-                     (float*)ws, g, M, CRS, slice_len, fast_a);
+                     (const int*)mt, (float*)ws, g, M, CRS, slice_len);
   long len = (long)CRS * K;
   hipLaunchKernelGGL(reduce_slabs_f32_kernel,
                      dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
                      dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
                      (float*)dw, nslices, len);
+}
+
+void cilfw_fill_mtable(void* mt, int M, int HoWo, int Wo, int stride,
+                       void* stream) {
+  hipLaunchKernelGGL(fill_mtable_kernel, dim3(cdiv(M, NTHREADS)),
+                     dim3(NTHREADS), 0, (hipStream_t)stream, (int*)mt, M,
+                     HoWo, Wo, stride);
 }
 
 int cilfw_conv2d_bwd_weight_nslices(int N, int C, int K, int R, int S,
