@@ -51,9 +51,9 @@ _PROTOS = {
     "cilfw_linear_fwd": [c_vp] * 4 + [c_i] * 3 + [c_vp],
     "cilfw_linear_dx": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_linear_dw": [c_vp] * 4 + [c_i] * 3 + [c_vp],
-    "cilfw_ce_fwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
+    "cilfw_ce_fwd": [c_vp] * 5 + [c_i, c_i, c_f, c_vp],
     "cilfw_ce_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
-    "cilfw_kd_fwd": [c_vp] * 5 + [c_i, c_i, c_f, c_vp],
+    "cilfw_kd_fwd": [c_vp] * 6 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
     "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_f, c_f, c_f, c_vp],
     "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
@@ -186,7 +186,9 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
     y = torch.empty_like(x)
     mean = torch.empty(C, dtype=torch.float32, device=x.device)
     invstd = torch.empty(C, dtype=torch.float32, device=x.device)
-    scratch = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+    gy = (M + 255) // 256  # keep in sync with rows_per_blk=256 in norm.hip
+    scratch = torch.empty((gy + 1) * 2 * C, dtype=torch.float32,
+                          device=x.device)
     gf = gamma.float().contiguous()
     bf = beta.float().contiguous()
     _lib.cilfw_bn_fwd(_ptr(x), _ptr(y), _ptr(gf), _ptr(bf),
@@ -203,14 +205,16 @@ def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training):
     C = x.shape[-1]
     M = x.numel() // C
     dx = torch.empty_like(x)
-    dgb = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+    gy = (M + 255) // 256  # keep in sync with rows_per_blk=256 in norm.hip
+    dgb = torch.empty((gy + 1) * 2 * C, dtype=torch.float32, device=x.device)
     gf = gamma.float().contiguous()
     _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(gf),
                       _ptr(mean), _ptr(invstd), _ptr(dgb),
                       c_l(M), c_i(C), c_i(1 if relu else 0),
                       c_i(1 if training else 0), _stream())
     _check("bn_bwd")
-    return dx, dgb[:C], dgb[C:]
+    base = gy * 2 * C
+    return dx, dgb[base:base + C], dgb[base + C:base + 2 * C]
 
 
 # ------------------------------------------------------------------ elementwise
@@ -333,8 +337,10 @@ def ce_fwd(logits, targets, smooth):
     M, C = logits.shape
     probs = torch.empty_like(logits)
     loss = torch.empty((), dtype=torch.float32, device=logits.device)
+    rowloss = torch.empty(M, dtype=torch.float32, device=logits.device)
     _lib.cilfw_ce_fwd(_ptr(logits), _ptr(targets.contiguous()), _ptr(probs),
-                      _ptr(loss), c_i(M), c_i(C), c_f(smooth), _stream())
+                      _ptr(loss), _ptr(rowloss), c_i(M), c_i(C), c_f(smooth),
+                      _stream())
     _check("ce_fwd")
     return loss, probs
 
@@ -356,8 +362,9 @@ def kd_fwd(s_logits, t_logits, T):
     ps = torch.empty_like(s)
     pt = torch.empty_like(s)
     loss = torch.empty((), dtype=torch.float32, device=s.device)
+    rowloss = torch.empty(M, dtype=torch.float32, device=s.device)
     _lib.cilfw_kd_fwd(_ptr(s), _ptr(t), _ptr(ps), _ptr(pt), _ptr(loss),
-                      c_i(M), c_i(C), c_f(T), _stream())
+                      _ptr(rowloss), c_i(M), c_i(C), c_f(T), _stream())
     _check("kd_fwd")
     return loss, ps, pt
 

@@ -14,7 +14,7 @@
 __global__ __launch_bounds__(NT)
 void ce_fwd_kernel(const float* __restrict__ logits,
                    const long* __restrict__ targets,
-                   float* __restrict__ probs, float* __restrict__ loss,
+                   float* __restrict__ probs, float* __restrict__ rowloss,
                    int M, int C, float smooth) {
   const int row = blockIdx.x;
   const float* lr = logits + (long)row * C;
@@ -52,9 +52,24 @@ void ce_fwd_kernel(const float* __restrict__ logits,
   }
   if (threadIdx.x == 0) {
     float nll = lse - lr[targets[row]];
-    float l = (1.f - smooth) * nll - smooth * (sum_logp / C);
-    atomicAdd(loss, l / M);
+    rowloss[row] = (1.f - smooth) * nll - smooth * (sum_logp / C);
   }
+}
+
+// deterministic mean over per-row losses (single block, fixed tree order)
+__global__ __launch_bounds__(NT)
+void loss_mean_kernel(const float* __restrict__ rowloss,
+                      float* __restrict__ loss, int M, float scale) {
+  __shared__ float red[NT];
+  float s = 0.f;
+  for (int i = threadIdx.x; i < M; i += NT) s += rowloss[i];
+  red[threadIdx.x] = s;
+  __syncthreads();
+  for (int o = NT / 2; o > 0; o >>= 1) {
+    if (threadIdx.x < o) red[threadIdx.x] += red[threadIdx.x + o];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) loss[0] = red[0] * scale;
 }
 
 // dlogits = (probs - (1-s)*onehot - s/C) * dloss / M  (fp32)
@@ -119,7 +134,7 @@ void kd_fwd_kernel(const float* __restrict__ s, const float* __restrict__ tt,
   __syncthreads();
   if (threadIdx.x == 0) {
     l = red[0] + red[1] + red[2] + red[3];
-    atomicAdd(loss, l * T * T / M);
+    loss[row] = l * T * T;  // per-row; loss_mean_kernel reduces
   }
 }
 
@@ -290,12 +305,15 @@ void herding_kernel(const float* __restrict__ f, const float* __restrict__ mu,
 extern "C" {
 
 void cilfw_ce_fwd(const void* logits, const void* targets, void* probs,
-                  void* loss, int M, int C, float smooth, void* stream) {
-  (void)hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
+                  void* loss, void* rowloss, int M, int C, float smooth,
+                  void* stream) {
   hipLaunchKernelGGL(ce_fwd_kernel, dim3(M), dim3(NT), 0,
                      (hipStream_t)stream, (const float*)logits,
-                     (const long*)targets, (float*)probs, (float*)loss, M, C,
-                     smooth);
+                     (const long*)targets, (float*)probs, (float*)rowloss, M,
+                     C, smooth);
+  hipLaunchKernelGGL(loss_mean_kernel, dim3(1), dim3(NT), 0,
+                     (hipStream_t)stream, (const float*)rowloss,
+                     (float*)loss, M, 1.f / M);
 }
 
 void cilfw_ce_bwd(const void* probs, const void* targets, const void* dloss,
@@ -308,11 +326,14 @@ void cilfw_ce_bwd(const void* probs, const void* targets, const void* dloss,
 }
 
 void cilfw_kd_fwd(const void* s, const void* t, void* ps, void* pt,
-                  void* loss, int M, int C, float T, void* stream) {
-  (void)hipMemsetAsync(loss, 0, sizeof(float), (hipStream_t)stream);
+                  void* loss, void* rowloss, int M, int C, float T,
+                  void* stream) {
   hipLaunchKernelGGL(kd_fwd_kernel, dim3(M), dim3(NT), 0,
                      (hipStream_t)stream, (const float*)s, (const float*)t,
-                     (float*)ps, (float*)pt, (float*)loss, M, C, T);
+                     (float*)ps, (float*)pt, (float*)rowloss, M, C, T);
+  hipLaunchKernelGGL(loss_mean_kernel, dim3(1), dim3(NT), 0,
+                     (hipStream_t)stream, (const float*)rowloss,
+                     (float*)loss, M, 1.f / M);
 }
 
 void cilfw_kd_bwd(const void* ps, const void* pt, const void* dloss, void* ds,

@@ -16,8 +16,11 @@
 // per-group bounds) over rows_per_blk rows, partials reduced in LDS, then one
 // fp32 atomic per (channel, quantity) per block.
 __global__ __launch_bounds__(NT)
-void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ sum,
-                    float* __restrict__ sumsq, long M, int C, int rows_per_blk) {
+void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ part,
+                    long M, int C, int rows_per_blk) {
+  // part: [gridDim.y][2][C] fp32 partials (deterministic tree reduce follows;
+  // the reference runs under cudnn.deterministic — cilfw matches: no fp32
+  // atomics anywhere in the training step)
   const int cg = threadIdx.x & 7;          // channel group (8 ch)
   const int rl = threadIdx.x >> 3;         // row lane 0..31
   const int c8 = blockIdx.x * 64 + cg * 8;
@@ -61,9 +64,19 @@ void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ sum,
     float acc = 0.f;
 #pragma unroll 8
     for (int r = 0; r < 32; ++r) acc += red[qi][r][cc];
-    atomicAdd(qi == 0 ? &sum[blockIdx.x * 64 + cc]
-                      : &sumsq[blockIdx.x * 64 + cc], acc);
+    part[((long)blockIdx.y * 2 + qi) * C + blockIdx.x * 64 + cc] = acc;
   }
+}
+
+// deterministic slab reduce: out[i] = sum_s ws[s][i]  (local copy — no RDC)
+__global__ __launch_bounds__(NT)
+void bn_reduce_slabs_kernel(const float* __restrict__ ws,
+                            float* __restrict__ out, int ns, long len) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= len) return;
+  float a = 0.f;
+  for (int s = 0; s < ns; ++s) a += ws[(long)s * len + i];
+  out[i] = a;
 }
 
 // pass 2: finalize mean/invstd (+ running stats update, training only)
@@ -140,7 +153,7 @@ void bn_bwd_sums_kernel(const bf16_t* __restrict__ dy,
                         const bf16_t* __restrict__ y,
                         const float* __restrict__ mean,
                         const float* __restrict__ invstd,
-                        float* __restrict__ dgamma, float* __restrict__ dbeta,
+                        float* __restrict__ part,
                         long M, int C, int rows_per_blk, int relu) {
   const int cg = threadIdx.x & 7;
   const int rl = threadIdx.x >> 3;
@@ -206,8 +219,7 @@ void bn_bwd_sums_kernel(const bf16_t* __restrict__ dy,
     float acc = 0.f;
 #pragma unroll 8
     for (int r = 0; r < 32; ++r) acc += red[qi][r][cc];
-    atomicAdd(qi == 0 ? &dgamma[blockIdx.x * 64 + cc]
-                      : &dbeta[blockIdx.x * 64 + cc], acc);
+    part[((long)blockIdx.y * 2 + qi) * C + blockIdx.x * 64 + cc] = acc;
   }
 }
 
@@ -423,13 +435,16 @@ void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
                   void* stream) {
   hipStream_t st = (hipStream_t)stream;
   if (training) {
-    (void)hipMemsetAsync(scratch_sums, 0, 2 * C * sizeof(float), st);
-    float* sum = (float*)scratch_sums;
-    float* sumsq = sum + C;
+    // scratch_sums: [gy][2][C] partials + [2][C] reduced (see wrapper sizing)
     int rows_per_blk = 256;
     dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
+    float* part = (float*)scratch_sums;
+    float* sum = part + (long)grid.y * 2 * C;
+    float* sumsq = sum + C;
     hipLaunchKernelGGL(bn_sums_kernel, grid, dim3(NT), 0, st,
-                       (const bf16_t*)x, sum, sumsq, M, C, rows_per_blk);
+                       (const bf16_t*)x, part, M, C, rows_per_blk);
+    hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT)),
+                       dim3(NT), 0, st, part, sum, (int)grid.y, (long)2 * C);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
                        st, sum, sumsq, (float*)mean, (float*)invstd,
                        (float*)running_mean, (float*)running_var, M, C,
@@ -452,17 +467,19 @@ void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                   const void* gamma, const void* mean, const void* invstd,
                   void* dgb, long M, int C, int relu,
                   int training, void* stream) {
-  // dgb: ONE fp32 buffer [dgamma | dbeta] (single memset, single allocation)
+  // dgb: [gy][2][C] partials followed by the reduced [dgamma | dbeta]
   hipStream_t st = (hipStream_t)stream;
-  float* dgamma = (float*)dgb;
-  float* dbeta = dgamma + C;
-  (void)hipMemsetAsync(dgb, 0, 2 * C * sizeof(float), st);
   int rows_per_blk = 256;
   dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
+  float* part = (float*)dgb;
+  float* dgamma = part + (long)grid.y * 2 * C;
+  float* dbeta = dgamma + C;
   hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
                      (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
-                     (const float*)mean, (const float*)invstd, dgamma,
-                     dbeta, M, C, rows_per_blk, relu);
+                     (const float*)mean, (const float*)invstd, part,
+                     M, C, rows_per_blk, relu);
+  hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT)), dim3(NT),
+                     0, st, part, dgamma, (int)grid.y, (long)2 * C);
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),

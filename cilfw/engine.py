@@ -25,6 +25,7 @@ from .models import CilModel
 from .cil import RehearsalMemory, save_task_checkpoint, load_task_checkpoint
 from .optim import FlatSGD, CosineLR
 from .utils.metrics import MetricLogger, SmoothedValue
+from .utils.trace import trace_range
 
 
 def init_seed(args):
@@ -85,6 +86,8 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
     for epoch in range(args.num_epochs):
         if train_sampler is not None:
             train_sampler.set_epoch(epoch)
+        epoch_range = trace_range(f"task{args.task_id}/epoch{epoch}")
+        epoch_range.__enter__()
         metric_logger = MetricLogger()
         metric_logger.meters["lr"] = SmoothedValue(fmt="{value:.6f}")
         t0 = time.time()
@@ -116,6 +119,7 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
                                  loss=loss.item())
             metric_logger.update_n(n=bs, acc1=accs[0])
             metric_logger.meters["lr"].update(optimizer.lr)
+        epoch_range.__exit__(None, None, None)
         metric_logger.synchronize_between_processes(device=torch.device(device))
         scheduler.step()
         ips = nimg * get_world_size() / max(time.time() - t0, 1e-9)
