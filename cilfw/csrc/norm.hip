@@ -68,15 +68,19 @@ void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ part,
   }
 }
 
-// deterministic slab reduce: out[i] = sum_s ws[s][i]  (local copy — no RDC)
+// deterministic slab reduce: out[i] = sum_s ws[s][i]. One WAVE per output
+// element — 64 lanes stride the slab axis in parallel, then a fixed-order
+// shfl tree (ns can be ~512: a thread-serial loop would be pure latency).
 __global__ __launch_bounds__(NT)
 void bn_reduce_slabs_kernel(const float* __restrict__ ws,
                             float* __restrict__ out, int ns, long len) {
-  long i = (long)blockIdx.x * NT + threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  long i = (long)blockIdx.x * (NT / WAVE) + (threadIdx.x >> 6);
   if (i >= len) return;
   float a = 0.f;
-  for (int s = 0; s < ns; ++s) a += ws[(long)s * len + i];
-  out[i] = a;
+  for (int s = lane; s < ns; s += WAVE) a += ws[(long)s * len + i];
+  for (int o = 32; o > 0; o >>= 1) a += __shfl_xor(a, o);
+  if (lane == 0) out[i] = a;
 }
 
 // pass 2: finalize mean/invstd (+ running stats update, training only)
@@ -443,8 +447,9 @@ void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
     float* sumsq = sum + C;
     hipLaunchKernelGGL(bn_sums_kernel, grid, dim3(NT), 0, st,
                        (const bf16_t*)x, part, M, C, rows_per_blk);
-    hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT)),
-                       dim3(NT), 0, st, part, sum, (int)grid.y, (long)2 * C);
+    hipLaunchKernelGGL(bn_reduce_slabs_kernel,
+                       dim3(cdiv(2 * C, NT / WAVE)), dim3(NT), 0, st, part,
+                       sum, (int)grid.y, (long)2 * C);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
                        st, sum, sumsq, (float*)mean, (float*)invstd,
                        (float*)running_mean, (float*)running_var, M, C,
@@ -478,8 +483,9 @@ void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                      (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
                      (const float*)mean, (const float*)invstd, part,
                      M, C, rows_per_blk, relu);
-  hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT)), dim3(NT),
-                     0, st, part, dgamma, (int)grid.y, (long)2 * C);
+  hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT / WAVE)),
+                     dim3(NT), 0, st, part, dgamma, (int)grid.y,
+                     (long)2 * C);
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),
