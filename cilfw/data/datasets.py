@@ -11,6 +11,11 @@ import pickle
 
 import numpy as np
 
+try:
+    from PIL import Image
+except ImportError:  # pragma: no cover
+    Image = None
+
 DATASET_STATS = {
     # mean/std in [0,1] per channel
     "cifar100": ((0.5071, 0.4865, 0.4409), (0.2673, 0.2564, 0.2762)),
@@ -53,6 +58,32 @@ def make_synthetic(num_classes=100, per_class=50, size=32, seed=0, channels=3):
     return x[perm], y[perm]
 
 
+def load_image(path):
+    """Decode one image file -> uint8 HWC RGB (lazy path-based datasets)."""
+    with Image.open(path) as im:
+        return np.asarray(im.convert("RGB"), dtype=np.uint8)
+
+
+def scan_imagefolder(root, split):
+    """ImageFolder layout (<root>/<split>/<class>/<img>) -> (paths object
+    ndarray, labels int64). Lazy: images decode at __getitem__ time, so the
+    CIL machinery (scenario splits, add_samples, rehearsal memory) moves only
+    path arrays — the reference's continuum ImageFolderDataset contract
+    (utils.py:171-185)."""
+    base = os.path.join(root, split)
+    classes = sorted(d for d in os.listdir(base)
+                     if os.path.isdir(os.path.join(base, d)))
+    paths, labels = [], []
+    for ci, cname in enumerate(classes):
+        cdir = os.path.join(base, cname)
+        for f in sorted(os.listdir(cdir)):
+            if f.lower().endswith((".jpg", ".jpeg", ".png", ".bmp", ".webp")):
+                paths.append(os.path.join(cdir, f))
+                labels.append(ci)
+    return (np.asarray(paths, dtype=object),
+            np.asarray(labels, dtype=np.int64), len(classes))
+
+
 def build_source(args, is_train):
     """-> (x uint8 (N,H,W,C), y int64 (N,), nb_classes, stats_key)."""
     name = args.data_set.lower()
@@ -74,6 +105,11 @@ def build_source(args, is_train):
         return x, y, nc, "synthetic"
     if name in ("imagenet100", "imagenet1000", "cub200"):
         nc = {"imagenet100": 100, "imagenet1000": 1000, "cub200": 200}[name]
+        split = "train" if is_train else "val"
+        if Image is not None and \
+                os.path.isdir(os.path.join(args.data_path, split)):
+            x, y, found = scan_imagefolder(args.data_path, split)
+            return x, y, found, "imagenet"
         per_class = 64 if is_train else 8  # synthetic stand-in (no network)
         x, y = make_synthetic(nc, per_class, args.input_size,
                               seed=0 if is_train else 1)

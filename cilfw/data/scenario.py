@@ -31,6 +31,9 @@ class TaskSet(Dataset):
 
     def __getitem__(self, i):
         img = self.x[i]
+        if isinstance(img, (str, bytes)) or (isinstance(img, np.str_)):
+            from .datasets import load_image
+            img = load_image(img)
         if self.transform is not None:
             img = self.transform(img)
         else:
