@@ -574,6 +574,20 @@ void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
 
 // ============================== launchers ==============================
 
+#include <stdlib.h>
+
+// BKT=64 halves barriers but its 55 KB LDS halves occupancy (2 vs 4 blocks/CU)
+// — the crossover is empirical, so the threshold is runtime-tunable.
+static int bk64_min_crs() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_BK64_MIN");
+    v = e ? atoi(e) : 512;
+    if (v <= 0) v = 1 << 30;  // 0 disables BKT=64
+  }
+  return v;
+}
+
 static int pick_ksplit(int nblocks, int nk) {
   // aim for >= 512 workgroups (2 per CU) without shredding the K loop
   int ks = 1;
@@ -591,7 +605,7 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int fast_a = (C % 16 == 0);
-  int use64 = (CRS >= 512);
+  int use64 = (CRS >= bk64_min_crs());
   int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(K, BN), ksplit);
   if (use64)
@@ -617,7 +631,7 @@ int cilfw_conv2d_fwd_ksplit(int N, int C, int K, int R, int S, int Ho,
                             int Wo) {
   int M = N * Ho * Wo;
   int CRS = C * R * S;
-  int nk = cdiv(CRS, CRS >= 512 ? 64 : 32);
+  int nk = cdiv(CRS, CRS >= bk64_min_crs() ? 64 : 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(K, BN), nk);
 }
 
@@ -629,7 +643,7 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);
-  int use64 = (RSK >= 512);
+  int use64 = (RSK >= bk64_min_crs());
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
   if (use64)
@@ -655,7 +669,7 @@ int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
                                  int S) {
   int M = N * H * W;
   int RSK = R * S * K;
-  int nk = cdiv(RSK, RSK >= 512 ? 64 : 32);
+  int nk = cdiv(RSK, RSK >= bk64_min_crs() ? 64 : 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(C, BN), nk);
 }
 
