@@ -605,7 +605,9 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int fast_a = (C % 16 == 0);
-  int use64 = (CRS >= bk64_min_crs());
+  // BKT=64 only when the grid is too small to exploit BKT=32's 2x occupancy
+  int use64 = (CRS >= bk64_min_crs()) &&
+              (cdiv(M, BM) * cdiv(K, BN) * ksplit < 768);
   int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(K, BN), ksplit);
   if (use64)
@@ -631,7 +633,7 @@ int cilfw_conv2d_fwd_ksplit(int N, int C, int K, int R, int S, int Ho,
                             int Wo) {
   int M = N * Ho * Wo;
   int CRS = C * R * S;
-  int nk = cdiv(CRS, CRS >= bk64_min_crs() ? 64 : 32);
+  int nk = cdiv(CRS, 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(K, BN), nk);
 }
 
@@ -643,7 +645,8 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);
-  int use64 = (RSK >= bk64_min_crs());
+  int use64 = (RSK >= bk64_min_crs()) &&
+              (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
   if (use64)
@@ -669,7 +672,7 @@ int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
                                  int S) {
   int M = N * H * W;
   int RSK = R * S * K;
-  int nk = cdiv(RSK, RSK >= bk64_min_crs() ? 64 : 32);
+  int nk = cdiv(RSK, 32);
   return pick_ksplit(cdiv(M, BM) * cdiv(C, BN), nk);
 }
 
