@@ -46,6 +46,8 @@ _PROTOS = {
     "cilfw_downsample_a_bwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
     "cilfw_gap_fwd": [c_vp] * 2 + [c_i] * 3 + [c_vp],
     "cilfw_gap_bwd": [c_vp] * 2 + [c_i] * 3 + [c_vp],
+    "cilfw_stride_gather": [c_vp] * 2 + [c_i] * 7 + [c_vp],
+    "cilfw_stride_scatter": [c_vp] * 2 + [c_i] * 7 + [c_vp],
     "cilfw_maxpool_fwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
     "cilfw_maxpool_bwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
     "cilfw_linear_fwd": [c_vp] * 4 + [c_i] * 3 + [c_vp],
@@ -125,6 +127,16 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
     N, Ho, Wo, K = dy.shape
     R, S, C, Kw = w.shape
     assert Kw == K
+    if R == 1 and S == 1 and stride > 1:
+        # strided 1x1 projection: 3/4 of positions are structurally zero —
+        # solve on the subsampled (stride-1) grid, then scatter
+        dxs = conv2d_bwd_data(dy, w, 1, 0, Ho, Wo)
+        dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
+        _lib.cilfw_stride_scatter(_ptr(dxs), _ptr(dx), c_i(N), c_i(H),
+                                  c_i(W_), c_i(C), c_i(stride), c_i(Ho),
+                                  c_i(Wo), _stream())
+        _check("stride_scatter")
+        return dx
     dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
     ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H, W_, C, K, R, S)
     ws = (torch.empty(ks * N * H * W_ * C, dtype=torch.float32,
@@ -161,6 +173,13 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
+    if R == 1 and S == 1 and stride > 1:
+        xg = torch.empty(N, Ho, Wo, C, dtype=torch.bfloat16, device=x.device)
+        _lib.cilfw_stride_gather(_ptr(x), _ptr(xg), c_i(N), c_i(H), c_i(W_),
+                                 c_i(C), c_i(stride), c_i(Ho), c_i(Wo),
+                                 _stream())
+        _check("stride_gather")
+        return conv2d_bwd_weight(dy, xg, 1, 0, R, S)
     mt = _mtable(N, Ho, Wo, stride, dy.device)
     dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
     ns = _lib.cilfw_conv2d_bwd_weight_nslices(N, C, K, R, S, Ho, Wo)

@@ -428,6 +428,44 @@ void maxpool_bwd_kernel(const bf16_t* __restrict__ dy,
   dx[i] = f2bf(acc);
 }
 
+// ---------------------- strided 1x1 helpers (gather/scatter subsampled grid)
+
+__global__ __launch_bounds__(NT)
+void stride_gather_kernel(const bf16_t* __restrict__ x,
+                          bf16_t* __restrict__ xg, int H, int W, int C,
+                          int s, int Ho, int Wo, long total_out) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_out) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int wo = (int)(rest % Wo);
+  rest /= Wo;
+  int ho = (int)(rest % Ho);
+  int n = (int)(rest / Ho);
+  xg[i] = x[(((long)n * H + ho * s) * W + wo * s) * C + c];
+}
+
+__global__ __launch_bounds__(NT)
+void stride_scatter_kernel(const bf16_t* __restrict__ dxs,
+                           bf16_t* __restrict__ dx, int H, int W, int C,
+                           int s, int Ho, int Wo, long total_in) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= total_in) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int w = (int)(rest % W);
+  rest /= W;
+  int h = (int)(rest % H);
+  int n = (int)(rest / H);
+  bf16_t v = 0;
+  if (h % s == 0 && w % s == 0) {
+    int ho = h / s, wo = w / s;
+    if (ho < Ho && wo < Wo)
+      v = dxs[(((long)n * Ho + ho) * Wo + wo) * C + c];
+  }
+  dx[i] = v;
+}
+
 // ============================== launchers ==============================
 
 extern "C" {
@@ -562,6 +600,24 @@ void cilfw_maxpool_bwd(const void* dy, const void* idx, void* dx, int N,
                      dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const int*)idx,
                      (bf16_t*)dx, N, H, W, C, kk, st, pad, Ho, Wo, total);
+}
+
+void cilfw_stride_gather(const void* x, void* xg, int N, int H, int W,
+                         int C, int s, int Ho, int Wo, void* stream) {
+  long total = (long)N * Ho * Wo * C;
+  hipLaunchKernelGGL(stride_gather_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)xg, H, W,
+                     C, s, Ho, Wo, total);
+}
+
+void cilfw_stride_scatter(const void* dxs, void* dx, int N, int H, int W,
+                          int C, int s, int Ho, int Wo, void* stream) {
+  long total = (long)N * H * W * C;
+  hipLaunchKernelGGL(stride_scatter_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)dxs, (bf16_t*)dx, H,
+                     W, C, s, Ho, Wo, total);
 }
 
 }  // extern "C"

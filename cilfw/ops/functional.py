@@ -58,19 +58,23 @@ class Conv2dNHWC(torch.autograd.Function):
         x, wc = ctx.saved_tensors
         stride, padding = ctx.stride, ctx.padding
         dy = dy.contiguous()
+        need_dx = ctx.needs_input_grad[0]  # stems skip the whole bwd-data pass
         if use_hip(dy):
-            dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1], x.shape[2])
+            dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1],
+                                       x.shape[2]) if need_dx else None
             dw = ext().conv2d_bwd_weight(dy, x, stride, padding,
                                          wc.shape[0], wc.shape[1])
             return dx, dw.to(ctx.w_dtype), None, None
         xf = _to_nchw(x).float()
         wf = wc.permute(3, 2, 0, 1).contiguous().float()
         dyf = _to_nchw(dy).float()
-        dxf = torch.nn.grad.conv2d_input(xf.shape, wf, dyf, stride=stride,
-                                         padding=padding)
+        dx = None
+        if need_dx:
+            dxf = torch.nn.grad.conv2d_input(xf.shape, wf, dyf, stride=stride,
+                                             padding=padding)
+            dx = _to_nhwc(dxf).to(x.dtype)
         dwf = torch.nn.grad.conv2d_weight(xf, wf.shape, dyf, stride=stride,
                                           padding=padding)
-        dx = _to_nhwc(dxf).to(x.dtype)
         dw = dwf.permute(2, 3, 1, 0).contiguous().to(ctx.w_dtype)  # (R,S,C,K)
         return dx, dw, None, None
 
