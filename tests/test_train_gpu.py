@@ -33,6 +33,34 @@ def test_gpu_two_task_run_learns():
 
 
 @pytest.mark.timeout(600)
+def test_gpu_resnet50_step():
+    """Bottleneck blocks + 7x7 stem + maxpool on the HIP path (ImageNet-style
+    geometry at reduced size)."""
+    from cilfw.models import CilModel
+    from cilfw import ops
+    from cilfw.distributed import DataParallelEngine
+    from cilfw.optim import FlatSGD
+    torch.manual_seed(0)
+    model = CilModel("resnet50", 224).to("cuda")  # 7x7/s2 stem + maxpool
+    model.prev_model_adaption(10)
+    model = model.to("cuda")
+    engine = DataParallelEngine(model)
+    opt = FlatSGD(engine, lr=0.1)
+    x = torch.randn(8, 64, 64, 3, device="cuda").to(torch.bfloat16)
+    y = torch.randint(0, 10, (8,), device="cuda")
+    for _ in range(2):
+        opt.zero_grad()
+        logits, feats = model(x)
+        loss = ops.cross_entropy(logits.float(), y)
+        loss.backward()
+        engine.finalize()
+        opt.step()
+    torch.cuda.synchronize()
+    assert feats.shape == (8, 2048)
+    assert torch.isfinite(loss)
+
+
+@pytest.mark.timeout(600)
 def test_gpu_determinism():
     """No fp32 atomics anywhere in the step: two identical runs bit-match."""
     a1 = run(_args(epochs=2))
