@@ -38,8 +38,8 @@ _PROTOS = {
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 12 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
-    "cilfw_bn_fwd": [c_vp] * 9 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
-    "cilfw_bn_bwd": [c_vp] * 8 + [c_l, c_i, c_i, c_i, c_vp],
+    "cilfw_bn_fwd": [c_vp] * 10 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
+    "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_downsample_a_fwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
@@ -197,8 +197,10 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
 # -------------------------------------------------------------------------- bn
 
 def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
-           relu):
+           relu, residual=None):
     _bf16(x, "bn_fwd.x")
+    if residual is not None:
+        _bf16(residual, "bn_fwd.residual")
     C = x.shape[-1]
     assert C % 8 == 0, "bn kernels vectorize over channels (C%8==0)"
     M = x.numel() // C
@@ -210,7 +212,7 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
                           device=x.device)
     gf = gamma.float().contiguous()
     bf = beta.float().contiguous()
-    _lib.cilfw_bn_fwd(_ptr(x), _ptr(y), _ptr(gf), _ptr(bf),
+    _lib.cilfw_bn_fwd(_ptr(x), _ptr(y), _ptr(residual), _ptr(gf), _ptr(bf),
                       _ptr(running_mean), _ptr(running_var), _ptr(mean),
                       _ptr(invstd), _ptr(scratch), c_l(M), c_i(C),
                       c_f(momentum), c_f(eps), c_i(1 if training else 0),
@@ -219,21 +221,22 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
     return y, mean, invstd
 
 
-def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training):
+def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training, want_dres=False):
     _bf16(dy, "bn_bwd.dy")
     C = x.shape[-1]
     M = x.numel() // C
     dx = torch.empty_like(x)
+    dres = torch.empty_like(x) if want_dres else None
     gy = (M + 255) // 256  # keep in sync with rows_per_blk=256 in norm.hip
     dgb = torch.empty((gy + 1) * 2 * C, dtype=torch.float32, device=x.device)
     gf = gamma.float().contiguous()
-    _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(gf),
-                      _ptr(mean), _ptr(invstd), _ptr(dgb),
+    _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(dres),
+                      _ptr(gf), _ptr(mean), _ptr(invstd), _ptr(dgb),
                       c_l(M), c_i(C), c_i(1 if relu else 0),
                       c_i(1 if training else 0), _stream())
     _check("bn_bwd")
     base = gy * 2 * C
-    return dx, dgb[base:base + C], dgb[base + C:base + 2 * C]
+    return dx, dgb[base:base + C], dgb[base + C:base + 2 * C], dres
 
 
 # ------------------------------------------------------------------ elementwise

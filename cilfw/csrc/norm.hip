@@ -111,6 +111,7 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
 // pass 3: y = gamma*(x-mean)*invstd + beta (+relu). 8 bf16 per thread.
 __global__ __launch_bounds__(NT)
 void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+                     const bf16_t* __restrict__ res,
                      const float* __restrict__ gamma,
                      const float* __restrict__ beta,
                      const float* __restrict__ mean,
@@ -130,6 +131,7 @@ void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
       int c = (int)(i % C);
       float v = (bf2f(x[i]) - params[2 * C + c]) * params[3 * C + c] *
                 params[c] + params[C + c];
+      if (res != nullptr) v += bf2f(res[i]);
       if (relu) v = fmaxf(v, 0.f);
       y[i] = f2bf(v);
     }
@@ -137,6 +139,9 @@ void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
   }
   int4 xv = *(const int4*)&x[i0];
   bf16_t* xe = (bf16_t*)&xv;
+  int4 rv;
+  const bf16_t* re = (const bf16_t*)&rv;
+  if (res != nullptr) rv = *(const int4*)&res[i0];
   bf16_t out[8];
   int c0 = (int)(i0 % C);  // C % 8 == 0 for all cilfw models
 #pragma unroll
@@ -144,6 +149,7 @@ void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
     int c = c0 + j;
     float v = (bf2f(xe[j]) - params[2 * C + c]) * params[3 * C + c] *
               params[c] + params[C + c];
+    if (res != nullptr) v += bf2f(re[j]);
     if (relu) v = fmaxf(v, 0.f);
     out[j] = f2bf(v);
   }
@@ -233,6 +239,7 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
                          const bf16_t* __restrict__ x,
                          const bf16_t* __restrict__ y,
                          bf16_t* __restrict__ dx,
+                         bf16_t* __restrict__ dres,
                          const float* __restrict__ gamma,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
@@ -255,6 +262,7 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
     int c = (int)(i % C);
     float g = bf2f(dy[i]);
     if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
+    if (dres != nullptr) dres[i] = f2bf(g);
     float v;
     if (training) {
       float xhat = (bf2f(x[i]) - params[C + c]) * params[2 * C + c];
@@ -470,7 +478,8 @@ void stride_scatter_kernel(const bf16_t* __restrict__ dxs,
 
 extern "C" {
 
-void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
+void cilfw_bn_fwd(const void* x, void* y, const void* res,
+                  const void* gamma, const void* beta,
                   void* running_mean, void* running_var, void* mean,
                   void* invstd, void* scratch_sums, long M, int C,
                   float momentum, float eps, int training, int relu,
@@ -502,13 +511,14 @@ void cilfw_bn_fwd(const void* x, void* y, const void* gamma, const void* beta,
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_apply_kernel, dim3((int)blocks), dim3(NT),
                      4 * C * sizeof(float), st, (const bf16_t*)x, (bf16_t*)y,
-                     (const float*)gamma, (const float*)beta,
-                     (const float*)mean, (const float*)invstd, total, C, relu);
+                     (const bf16_t*)res, (const float*)gamma,
+                     (const float*)beta, (const float*)mean,
+                     (const float*)invstd, total, C, relu);
 }
 
 void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
-                  const void* gamma, const void* mean, const void* invstd,
-                  void* dgb, long M, int C, int relu,
+                  void* dres, const void* gamma, const void* mean,
+                  const void* invstd, void* dgb, long M, int C, int relu,
                   int training, void* stream) {
   // dgb: [gy][2][C] partials followed by the reduced [dgamma | dbeta]
   hipStream_t st = (hipStream_t)stream;
@@ -529,7 +539,7 @@ void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),
                      5 * C * sizeof(float), st, (const bf16_t*)dy,
                      (const bf16_t*)x, (const bf16_t*)y, (bf16_t*)dx,
-                     (const float*)gamma, (const float*)mean,
+                     (bf16_t*)dres, (const float*)gamma, (const float*)mean,
                      (const float*)invstd, (const float*)dgamma,
                      (const float*)dbeta, total, M, C, relu, training);
 }

@@ -43,7 +43,13 @@ class BatchNormAct2d(nn.Module):
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
+        if residual is not None:
+            # fused y = relu(bn(x) + residual) — residual-block tail
+            return CF.batchnorm_add_relu(x, residual, self.weight, self.bias,
+                                         self.running_mean, self.running_var,
+                                         self.momentum, self.eps,
+                                         self.training)
         return CF.batchnorm_act(x, self.weight, self.bias, self.running_mean,
                                 self.running_var, self.momentum, self.eps,
                                 self.training, self.relu)

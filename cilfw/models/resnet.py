@@ -29,9 +29,9 @@ class BasicBlockB(nn.Module):
             self.proj = None
 
     def forward(self, x):
-        out = self.bn2(self.conv2(self.bn1(self.conv1(x))))
+        out = self.bn1(self.conv1(x))
         residual = self.proj_bn(self.proj(x)) if self.proj is not None else x
-        return CF.add_relu(out, residual)
+        return self.bn2(self.conv2(out), residual)  # fused bn+add+relu
 
 
 class Bottleneck(nn.Module):
@@ -55,9 +55,8 @@ class Bottleneck(nn.Module):
     def forward(self, x):
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
         residual = self.proj_bn(self.proj(x)) if self.proj is not None else x
-        return CF.add_relu(out, residual)
+        return self.bn3(self.conv3(out), residual)  # fused bn+add+relu
 
 
 class ResNet(nn.Module):

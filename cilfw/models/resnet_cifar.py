@@ -33,9 +33,8 @@ class BasicBlockA(nn.Module):
 
     def forward(self, x):
         out = self.bn_a(self.conv_a(x))
-        out = self.bn_b(self.conv_b(out))
         residual = CF.downsample_a(x) if self.downsample else x
-        return CF.add_relu(out, residual)
+        return self.bn_b(self.conv_b(out), residual)  # fused bn+add+relu
 
 
 class CifarResNet(nn.Module):

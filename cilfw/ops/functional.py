@@ -133,8 +133,8 @@ class BatchNormAct(torch.autograd.Function):
         relu = ctx.relu
         dy = dy.contiguous()
         if use_hip(dy):
-            dx, dgamma, dbeta = ext().bn_bwd(dy, x, gamma, mean, invstd, y,
-                                             relu, ctx.training)
+            dx, dgamma, dbeta, _ = ext().bn_bwd(dy, x, gamma, mean, invstd,
+                                                y, relu, ctx.training)
             return (dx, dgamma, dbeta) + (None,) * 6
         C = x.shape[-1]
         dyf = dy.float().reshape(-1, C)
@@ -158,6 +158,74 @@ def batchnorm_act(x, gamma, beta, running_mean, running_var, momentum=0.1, eps=1
                   training=True, relu=False):
     return BatchNormAct.apply(x, gamma, beta, running_mean, running_var, momentum,
                               eps, training, relu)
+
+
+class BatchNormAddReLU(torch.autograd.Function):
+    """y = relu(bn(x) + residual) — the residual-block tail fused into the BN
+    apply/backward kernels (saves the add_relu round-trips; dres == the
+    relu-masked dy the BN backward already computes)."""
+
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
+                momentum, eps, training):
+        ctx.eps = eps
+        if use_hip(x):
+            y, save_mean, save_invstd = ext().bn_fwd(
+                x, gamma, beta, running_mean, running_var, momentum, eps,
+                training, True, residual=residual)
+            ctx.training = training
+            ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+            return y
+        N, H, W, C = x.shape
+        xf = x.float().reshape(-1, C)
+        if training:
+            mean = xf.mean(dim=0)
+            var = xf.var(dim=0, unbiased=False)
+            m = xf.shape[0]
+            with torch.no_grad():
+                running_mean.mul_(1 - momentum).add_(momentum * mean)
+                unbiased = var * (m / max(m - 1, 1))
+                running_var.mul_(1 - momentum).add_(momentum * unbiased)
+        else:
+            mean, var = running_mean.float(), running_var.float()
+        invstd = torch.rsqrt(var + eps)
+        yf = (xf - mean) * invstd * gamma.float() + beta.float()
+        yf = (yf + residual.float().reshape(-1, C)).clamp_min(0)
+        y = yf.reshape(N, H, W, C).to(x.dtype)
+        ctx.training = training
+        ctx.save_for_backward(x, gamma, mean, invstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, invstd, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_hip(dy):
+            dx, dgamma, dbeta, dres = ext().bn_bwd(
+                dy, x, gamma, mean, invstd, y, True, ctx.training,
+                want_dres=True)
+            return (dx, dres, dgamma, dbeta) + (None,) * 5
+        C = x.shape[-1]
+        dyf = dy.float().reshape(-1, C) * (y.float().reshape(-1, C) > 0)
+        xf = x.float().reshape(-1, C)
+        xhat = (xf - mean) * invstd
+        dgamma = (dyf * xhat).sum(dim=0)
+        dbeta = dyf.sum(dim=0)
+        if ctx.training:
+            m = xf.shape[0]
+            dxf = (gamma.float() * invstd / m) * (
+                m * dyf - dbeta - xhat * dgamma)
+        else:
+            dxf = dyf * gamma.float() * invstd
+        dres = dyf.reshape_as(x).to(dy.dtype)
+        return (dxf.reshape_as(x).to(x.dtype), dres, dgamma.to(gamma.dtype),
+                dbeta.to(gamma.dtype)) + (None,) * 5
+
+
+def batchnorm_add_relu(x, residual, gamma, beta, running_mean, running_var,
+                       momentum=0.1, eps=1e-5, training=True):
+    return BatchNormAddReLU.apply(x, residual, gamma, beta, running_mean,
+                                  running_var, momentum, eps, training)
 
 
 # ------------------------------------------------------------------------- add+ReLU

@@ -178,3 +178,29 @@ def test_accuracy_topk():
     a1, a2 = ops.accuracy(logits, targets, topk=(1, 2))
     assert abs(a1 - 200.0 / 3) < 1e-6
     assert abs(a2 - 100.0) < 1e-6
+
+
+def test_bn_add_relu_fused_matches_composite():
+    torch.manual_seed(2)
+    C = 8
+    x = torch.randn(4, 5, 5, C, requires_grad=True)
+    res = torch.randn(4, 5, 5, C, requires_grad=True)
+    g = (torch.rand(C) + 0.5).requires_grad_()
+    b = torch.randn(C, requires_grad=True)
+    rm, rv = torch.zeros(C), torch.ones(C)
+    y = ops.batchnorm_add_relu(x, res, g, b, rm, rv, training=True)
+    # composite oracle
+    x2 = x.detach().requires_grad_()
+    res2 = res.detach().requires_grad_()
+    g2 = g.detach().requires_grad_()
+    b2 = b.detach().requires_grad_()
+    rm2, rv2 = torch.zeros(C), torch.ones(C)
+    y2 = ops.add_relu(ops.batchnorm_act(x2, g2, b2, rm2, rv2, training=True),
+                      res2)
+    assert torch.allclose(y, y2, atol=1e-5)
+    assert torch.allclose(rm, rm2, atol=1e-7)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    y2.backward(dy)
+    for a, c in [(x, x2), (res, res2), (g, g2), (b, b2)]:
+        assert torch.allclose(a.grad, c.grad, atol=1e-5)

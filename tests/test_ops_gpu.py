@@ -281,3 +281,26 @@ def test_resnet20_step_matches_cpu():
     ghead_g = mg.fc.heads[0].weight.grad
     ghead_c = mc.fc.heads[0].weight.grad
     _cmp(ghead_g, ghead_c, rtol=0.05, atol=0.05, what="head dw")
+
+
+def test_bn_add_relu_gpu():
+    C = 64
+    xg, xc = _pair(4, 8, 8, C, seed=30)
+    rg, rc = _pair(4, 8, 8, C, seed=31)
+    xg, xc = xg.requires_grad_(), xc.requires_grad_()
+    rg, rc = rg.requires_grad_(), rc.requires_grad_()
+    g = torch.rand(C) + 0.5
+    b = torch.randn(C)
+    gg, gc = g.clone().cuda().requires_grad_(), g.clone().requires_grad_()
+    bg, bc = b.clone().cuda().requires_grad_(), b.clone().requires_grad_()
+    yg = CF.batchnorm_add_relu(xg, rg, gg, bg, torch.zeros(C).cuda(),
+                               torch.ones(C).cuda(), training=True)
+    yc = CF.batchnorm_add_relu(xc, rc, gc, bc, torch.zeros(C),
+                               torch.ones(C), training=True)
+    _cmp(yg, yc, what="bn_add_relu fwd")
+    dy = torch.randn(yc.shape).to(torch.bfloat16)
+    yg.backward(dy.cuda())
+    yc.backward(dy)
+    _cmp(xg.grad, xc.grad, rtol=0.03, atol=0.03, what="bn_add_relu dx")
+    _cmp(rg.grad, rc.grad, what="bn_add_relu dres")
+    _cmp(gg.grad, gc.grad, what="bn_add_relu dgamma")
