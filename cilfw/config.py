@@ -90,6 +90,10 @@ def get_args_parser():
     parser.add_argument("--no_aug", action="store_true", default=False)
     parser.add_argument("--ddp_bucket_mb", default=25.0, type=float,
                         help="gradient all-reduce bucket size (MB)")
+    parser.add_argument("--gpu_data", action="store_true", default=False,
+                        help="HBM-resident task data + on-device batch "
+                             "assembly (crop/flip/normalize) — bypasses the "
+                             "Python DataLoader for array-backed datasets")
     parser.add_argument("--compat_step_barrier", action="store_true", default=False,
                         help="reproduce the reference's per-training-step "
                              "dist.barrier (template.py:272) — a perf bug kept "

@@ -20,6 +20,7 @@ def build_dataset(is_train, args, transform="auto"):
     """(scenario, nb_classes) — the reference's build_dataset contract
     (utils.py:188-207, called template.py:203-204)."""
     x, y, nb_classes, stats_key = build_source(args, is_train)
+    args._stats_key = stats_key
     if transform == "auto":
         transform = (TrainTransform(args, stats_key) if is_train
                      else EvalTransform(args, stats_key))

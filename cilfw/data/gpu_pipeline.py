@@ -1,0 +1,102 @@
+"""Device-resident data pipeline — feeds the HIP training step at full rate.
+
+A Python-side DataLoader tops out far below the ~20k imgs/s the MI355X step
+consumes at CIFAR scale. When the task's images are array-backed (CIFAR /
+synthetic; not lazy ImageFolder paths), cilfw uploads the task set ONCE as
+uint8 (CIFAR-100 full train split = 150 MB — nothing against 288 GB HBM) and
+assembles every batch on-device: gather -> random crop(+pad) -> horizontal
+flip -> normalize -> bf16, all torch index ops on the GPU.
+
+Sharding reproduces DistributedSampler semantics exactly: same seed+epoch
+permutation on every rank, pad-by-repetition to a multiple of world size,
+rank-strided slice (cilfw/data/sampler.py contract).
+"""
+
+import numpy as np
+import torch
+
+
+class GpuTaskLoader:
+    def __init__(self, taskset, batch_size, device, mean, std, world=1, rank=0,
+                 shuffle=True, seed=0, augment=True, drop_last=True,
+                 dtype=torch.bfloat16, pad=4):
+        assert taskset.x.dtype == np.uint8, \
+            "GpuTaskLoader needs array-backed images (not lazy paths)"
+        self.images = torch.from_numpy(np.ascontiguousarray(taskset.x)).to(
+            device)
+        self.labels = torch.from_numpy(np.ascontiguousarray(
+            taskset.y)).to(device)
+        self.batch_size = batch_size
+        self.device = device
+        self.world, self.rank = world, rank
+        self.shuffle, self.seed = shuffle, seed
+        self.augment = augment
+        self.drop_last = drop_last
+        self.dtype = dtype
+        self.pad = pad
+        self.epoch = 0
+        self.mean = (torch.tensor(mean, device=device).view(1, 1, 1, -1)
+                     * 255.0)
+        self.std = (torch.tensor(std, device=device).view(1, 1, 1, -1)
+                    * 255.0)
+        n = len(self.labels)
+        import math
+        if drop_last and n % world != 0:
+            self.num_samples = n // world
+        else:
+            self.num_samples = math.ceil(n / world)
+
+    def set_epoch(self, epoch):
+        self.epoch = epoch
+
+    def __len__(self):
+        if self.drop_last:
+            return self.num_samples // self.batch_size
+        return (self.num_samples + self.batch_size - 1) // self.batch_size
+
+    def _shard_indices(self):
+        n = len(self.labels)
+        if self.shuffle:
+            g = torch.Generator()
+            g.manual_seed(self.seed + self.epoch)
+            idx = torch.randperm(n, generator=g)
+        else:
+            idx = torch.arange(n)
+        total = self.num_samples * self.world
+        if not self.drop_last and total > n:
+            reps = (total - n + n - 1) // n
+            idx = torch.cat([idx] + [idx] * reps)[:total]
+        else:
+            idx = idx[:total]
+        shard = idx[self.rank:total:self.world]
+        return shard.to(self.device)
+
+    def __iter__(self):
+        shard = self._shard_indices()
+        g = torch.Generator(device=self.device)
+        g.manual_seed(self.seed * 1000003 + self.epoch * 131 + self.rank)
+        nb = len(self)
+        for b in range(nb):
+            idx = shard[b * self.batch_size:(b + 1) * self.batch_size]
+            imgs = self.images[idx].float()
+            if self.augment:
+                imgs = self._augment(imgs, g)
+            imgs = ((imgs - self.mean) / self.std).to(self.dtype)
+            yield imgs, self.labels[idx], None
+
+    def _augment(self, imgs, g):
+        N, H, W, C = imgs.shape
+        p = self.pad
+        padded = torch.zeros(N, H + 2 * p, W + 2 * p, C, device=self.device)
+        padded[:, p:p + H, p:p + W] = imgs
+        oy = torch.randint(0, 2 * p + 1, (N,), device=self.device,
+                           generator=g)
+        ox = torch.randint(0, 2 * p + 1, (N,), device=self.device,
+                           generator=g)
+        rows = oy.view(N, 1) + torch.arange(H, device=self.device)
+        cols = ox.view(N, 1) + torch.arange(W, device=self.device)
+        imgs = padded[torch.arange(N, device=self.device).view(N, 1, 1),
+                      rows.view(N, H, 1), cols.view(N, 1, W)]
+        flip = torch.rand(N, device=self.device, generator=g) < 0.5
+        imgs[flip] = imgs[flip].flip(2)
+        return imgs

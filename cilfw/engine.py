@@ -17,7 +17,8 @@ import torch
 from torch.utils.data import DataLoader
 
 from . import ops
-from .data import build_dataset, DistributedSampler
+from .data import build_dataset, DistributedSampler, DATASET_STATS
+from .data.gpu_pipeline import GpuTaskLoader
 from .data.transforms import EvalTransform
 from .distributed import (init_distributed_mode, DataParallelEngine, barrier,
                           get_world_size, get_rank)
@@ -86,6 +87,8 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
     for epoch in range(args.num_epochs):
         if train_sampler is not None:
             train_sampler.set_epoch(epoch)
+        elif hasattr(train_loader, "set_epoch"):
+            train_loader.set_epoch(epoch)
         epoch_range = trace_range(f"task{args.task_id}/epoch{epoch}")
         epoch_range.__enter__()
         metric_logger = MetricLogger()
@@ -137,8 +140,16 @@ def extract_task_features(model, dataset, device, args):
     deterministic so the resulting memory is rank-identical."""
     model.eval()
     dtype = compute_dtype(args)
-    loader = DataLoader(dataset, batch_size=args.batch_size, shuffle=False,
-                        num_workers=args.workers, drop_last=False)
+    if (args.gpu_data and str(device).startswith("cuda")
+            and dataset.x.dtype == np.uint8):
+        mean, std = DATASET_STATS[getattr(args, "_stats_key", "synthetic")]
+        loader = GpuTaskLoader(dataset, args.batch_size, device, mean, std,
+                               shuffle=False, augment=False, drop_last=False,
+                               dtype=dtype)
+    else:
+        loader = DataLoader(dataset, batch_size=args.batch_size,
+                            shuffle=False, num_workers=args.workers,
+                            drop_last=False)
     feats = []
     for inputs, _targets, _tids in loader:
         inputs = inputs.to(device, non_blocking=True).to(dtype)
@@ -190,14 +201,32 @@ def run(args):
             dataset_train.add_samples(mx, my, mt)
 
         world, rank = get_world_size(), get_rank()
-        train_sampler = DistributedSampler(dataset_train, world, rank,
-                                           shuffle=True, seed=args.seed)
-        val_sampler = DistributedSampler(dataset_val, world, rank, shuffle=False)
-        train_loader = DataLoader(dataset_train, batch_size=args.batch_size,
-                                  sampler=train_sampler, num_workers=args.workers,
-                                  drop_last=True, persistent_workers=args.workers > 0)
-        val_loader = DataLoader(dataset_val, batch_size=args.batch_size,
-                                sampler=val_sampler, num_workers=args.workers)
+        use_gpu_data = (args.gpu_data and str(device).startswith("cuda")
+                        and dataset_train.x.dtype == np.uint8)
+        if use_gpu_data:
+            mean, std = DATASET_STATS[getattr(args, "_stats_key", "synthetic")]
+            train_sampler = None
+            train_loader = GpuTaskLoader(
+                dataset_train, args.batch_size, device, mean, std,
+                world=world, rank=rank, shuffle=True, seed=args.seed,
+                augment=not args.no_aug, drop_last=True,
+                dtype=compute_dtype(args))
+            val_loader = GpuTaskLoader(
+                dataset_val, args.batch_size, device, mean, std,
+                world=world, rank=rank, shuffle=False, augment=False,
+                drop_last=False, dtype=compute_dtype(args))
+        else:
+            train_sampler = DistributedSampler(dataset_train, world, rank,
+                                               shuffle=True, seed=args.seed)
+            val_sampler = DistributedSampler(dataset_val, world, rank,
+                                             shuffle=False)
+            train_loader = DataLoader(
+                dataset_train, batch_size=args.batch_size,
+                sampler=train_sampler, num_workers=args.workers,
+                drop_last=True, persistent_workers=args.workers > 0)
+            val_loader = DataLoader(dataset_val, batch_size=args.batch_size,
+                                    sampler=val_sampler,
+                                    num_workers=args.workers)
 
         model.prev_model_adaption(args.increment_per_task)
         engine = DataParallelEngine(model, bucket_mb=args.ddp_bucket_mb)

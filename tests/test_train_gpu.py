@@ -33,6 +33,16 @@ def test_gpu_two_task_run_learns():
 
 
 @pytest.mark.timeout(600)
+def test_gpu_data_pipeline_run():
+    """HBM-resident data path (--gpu_data) end-to-end."""
+    args = _args(epochs=3)
+    args.gpu_data = True
+    accs = run(args)
+    assert len(accs) == 2
+    assert accs[0] > 40.0, f"--gpu_data run failed to learn: {accs}"
+
+
+@pytest.mark.timeout(600)
 def test_gpu_resnet50_step():
     """Bottleneck blocks + 7x7 stem + maxpool on the HIP path (ImageNet-style
     geometry at reduced size)."""
