@@ -94,6 +94,9 @@ def get_args_parser():
                         help="HBM-resident task data + on-device batch "
                              "assembly (crop/flip/normalize) — bypasses the "
                              "Python DataLoader for array-backed datasets")
+    parser.add_argument("--no_step_graph", action="store_true", default=False,
+                        help="disable hipGraph capture of the training step "
+                             "(graphs are on by default with --gpu_data)")
     parser.add_argument("--compat_step_barrier", action="store_true", default=False,
                         help="reproduce the reference's per-training-step "
                              "dist.barrier (template.py:272) — a perf bug kept "

@@ -74,6 +74,30 @@ def evaluate(model, loader, device, args, header="Test:"):
     return acc1
 
 
+class _GraphedStep:
+    """hipGraph-captured training step for the in-engine hot loop: the whole
+    student fwd + teacher fwd + CE/KD + backward + fused SGD replays as one
+    graph launch. Batches are copied into static input buffers; outputs
+    (losses, logits) are static storages read after each replay. Re-captured
+    per epoch (the cosine LR is baked into the captured SGD launch)."""
+
+    def __init__(self, step_fn, x0, y0):
+        # capture EXECUTES one real step on (x0, y0) — the caller uses
+        # self.out for that batch and must not replay it again (so no batch
+        # is ever stepped twice; prior eager steps have warmed the allocator)
+        self.static_x = x0.clone()
+        self.static_y = y0.clone()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.out = step_fn(self.static_x, self.static_y)
+
+    def __call__(self, x, y):
+        self.static_x.copy_(x)
+        self.static_y.copy_(y)
+        self.graph.replay()
+        return self.out
+
+
 def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
                    train_sampler, val_loader, device, args):
     dtype = compute_dtype(args)
@@ -83,6 +107,27 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         # n/(n+m) scaling the reference documented but never wired
         # (README.md:175-176 / dead flag template.py:48)
         lambda_kd = known / (known + args.increment_per_task)
+    def step_fn(inputs, targets):
+        optimizer.zero_grad()
+        logits, _features = model(inputs)
+        loss_ce = ops.cross_entropy(logits.float(), targets, args.smooth)
+        if teacher is not None:
+            with torch.no_grad():
+                t_logits, _ = teacher(inputs)
+            loss_kd = ops.kd_loss(logits[:, :known].float(),
+                                  t_logits.float(), args.kd_temperature)
+            loss = loss_ce + lambda_kd * loss_kd
+        else:
+            loss_kd = torch.zeros((), device=logits.device)
+            loss = loss_ce
+        loss.backward()
+        engine.finalize()
+        optimizer.step()
+        return logits, loss_ce, loss_kd, loss
+
+    can_graph = (str(device).startswith("cuda") and not args.no_step_graph
+                 and not args.compat_step_barrier
+                 and not isinstance(train_loader, DataLoader))
     model.train()
     for epoch in range(args.num_epochs):
         if train_sampler is not None:
@@ -95,25 +140,27 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         metric_logger.meters["lr"] = SmoothedValue(fmt="{value:.6f}")
         t0 = time.time()
         nimg = 0
+        graphed = None  # per-epoch: the cosine LR is baked into the capture
+        first_of_epoch = True
         for inputs, targets, _tids in train_loader:
             inputs, targets = _to_device(inputs, targets, device, dtype)
-            optimizer.zero_grad()
-            logits, _features = model(inputs)
-            loss_ce = ops.cross_entropy(logits.float(), targets, args.smooth)
-            if teacher is not None:
-                with torch.no_grad():
-                    t_logits, _ = teacher(inputs)
-                loss_kd = ops.kd_loss(logits[:, :known].float(),
-                                      t_logits.float(), args.kd_temperature)
-                loss = loss_ce + lambda_kd * loss_kd
+            if can_graph and graphed is None and not first_of_epoch:
+                # capture on this batch; the capture run IS its training step
+                try:
+                    graphed = _GraphedStep(step_fn, inputs, targets)
+                    logits, loss_ce, loss_kd, loss = graphed.out
+                except Exception as e:
+                    print(f"[engine] step-graph capture failed "
+                          f"({type(e).__name__}: {e}); running eager")
+                    can_graph = False
+                    logits, loss_ce, loss_kd, loss = step_fn(inputs, targets)
+            elif graphed is not None:
+                logits, loss_ce, loss_kd, loss = graphed(inputs, targets)
             else:
-                loss_kd = torch.zeros((), device=logits.device)
-                loss = loss_ce
-            loss.backward()
-            engine.finalize()
-            optimizer.step()
-            if args.compat_step_barrier:
-                barrier()  # reference's per-step barrier (template.py:272)
+                logits, loss_ce, loss_kd, loss = step_fn(inputs, targets)
+                if args.compat_step_barrier:
+                    barrier()  # reference per-step barrier (template.py:272)
+            first_of_epoch = False
             accs = ops.accuracy(logits, targets,
                                 topk=(1, min(5, logits.shape[1])))
             bs = targets.shape[0]
