@@ -41,6 +41,3 @@ DEV FragIdx frag_idx() {
   int lane = threadIdx.x & (WAVE - 1);
   return {lane & 15, lane >> 4};
 }
-
-// atomicAdd on float — fine on gfx950 (HBM-backed fp32 atomics)
-DEV void atomic_add_f32(float* p, float v) { atomicAdd(p, v); }

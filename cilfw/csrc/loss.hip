@@ -8,8 +8,9 @@
 #define NT 256
 
 // ---------------------------------------------- fused CE (+label smoothing) fwd
-// logits fp32 (M, C); writes probs (M, C) fp32 and atomically accumulates the
-// mean loss into loss[0] (pre-zeroed). One block per row.
+// logits fp32 (M, C); writes probs (M, C) fp32 and per-row losses (reduced to
+// the scalar mean by loss_mean_kernel in a fixed order — deterministic).
+// One block per row.
 
 __global__ __launch_bounds__(NT)
 void ce_fwd_kernel(const float* __restrict__ logits,

@@ -225,7 +225,8 @@ def _center_crop(img, size):
 
 
 def color_jitter(img, strength):
-    img = _brightness(img, 0) if strength <= 0 else img
+    if strength <= 0:
+        return img
     for fn in (_brightness, _contrast, _color):
         f = 1.0 + np.random.uniform(-strength, strength)
         base = {_brightness: np.zeros_like(img),
