@@ -79,13 +79,25 @@ def main():
                     dtype=dtype)
     y = torch.randint(0, 100, (B,), device=device)
 
+    # the frozen teacher's forward is independent of the student's: run it on
+    # a side HIP stream, overlapped with the student forward (fork/join — also
+    # captured as a forked hipGraph)
+    tstream = torch.cuda.Stream() if (use_cuda and teacher is not None) else None
+
     def step():
         opt.zero_grad()
+        if tstream is not None:
+            tstream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(tstream), torch.no_grad():
+                t_logits, _ = teacher(x)
         logits, _ = model(x)
         loss = ops.cross_entropy(logits.float(), y)
         if teacher is not None:
-            with torch.no_grad():
-                t_logits, _ = teacher(x)
+            if tstream is not None:
+                torch.cuda.current_stream().wait_stream(tstream)
+            else:
+                with torch.no_grad():
+                    t_logits, _ = teacher(x)
             loss = loss + 0.5 * ops.kd_loss(logits[:, :known].float(),
                                             t_logits.float(), 2.0)
         loss.backward()

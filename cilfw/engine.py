@@ -107,13 +107,26 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         # n/(n+m) scaling the reference documented but never wired
         # (README.md:175-176 / dead flag template.py:48)
         lambda_kd = known / (known + args.increment_per_task)
+    # frozen-teacher forward overlaps the student forward on a side HIP stream
+    # (independent until the KD loss joins them; capture-compatible fork/join)
+    tstream = (torch.cuda.Stream()
+               if (teacher is not None and str(device).startswith("cuda"))
+               else None)
+
     def step_fn(inputs, targets):
         optimizer.zero_grad()
+        if tstream is not None:
+            tstream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(tstream), torch.no_grad():
+                t_logits, _ = teacher(inputs)
         logits, _features = model(inputs)
         loss_ce = ops.cross_entropy(logits.float(), targets, args.smooth)
         if teacher is not None:
-            with torch.no_grad():
-                t_logits, _ = teacher(inputs)
+            if tstream is not None:
+                torch.cuda.current_stream().wait_stream(tstream)
+            else:
+                with torch.no_grad():
+                    t_logits, _ = teacher(inputs)
             loss_kd = ops.kd_loss(logits[:, :known].float(),
                                   t_logits.float(), args.kd_temperature)
             loss = loss_ce + lambda_kd * loss_kd
