@@ -81,12 +81,23 @@ class _GraphedStep:
     (losses, logits) are static storages read after each replay. Re-captured
     per epoch (the cosine LR is baked into the captured SGD launch)."""
 
-    def __init__(self, step_fn, x0, y0):
-        # capture EXECUTES one real step on (x0, y0) — the caller uses
-        # self.out for that batch and must not replay it again (so no batch
-        # is ever stepped twice; prior eager steps have warmed the allocator)
+    def __init__(self, step_fn, x0, y0, model):
+        # torch.cuda.graph REQUIRES a side-stream warmup before capture. The
+        # warmup runs fwd+bwd WITHOUT the optimizer step and with BN running
+        # stats snapshot/restored, so it leaves no trace on training state;
+        # the capture itself then EXECUTES one real step on (x0, y0) — the
+        # caller uses self.out for that batch and must not replay it again.
         self.static_x = x0.clone()
         self.static_y = y0.clone()
+        stats = [(b, b.clone()) for n, b in model.named_buffers()
+                 if "running" in n]
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            step_fn(self.static_x, self.static_y, update=False)
+        torch.cuda.current_stream().wait_stream(s)
+        for b, saved in stats:
+            b.copy_(saved)
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self.out = step_fn(self.static_x, self.static_y)
@@ -113,7 +124,7 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
                if (teacher is not None and str(device).startswith("cuda"))
                else None)
 
-    def step_fn(inputs, targets):
+    def step_fn(inputs, targets, update=True):
         optimizer.zero_grad()
         if tstream is not None:
             tstream.wait_stream(torch.cuda.current_stream())
@@ -135,7 +146,8 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
             loss = loss_ce
         loss.backward()
         engine.finalize()
-        optimizer.step()
+        if update:
+            optimizer.step()
         return logits, loss_ce, loss_kd, loss
 
     can_graph = (str(device).startswith("cuda") and not args.no_step_graph
@@ -160,7 +172,7 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
             if can_graph and graphed is None and not first_of_epoch:
                 # capture on this batch; the capture run IS its training step
                 try:
-                    graphed = _GraphedStep(step_fn, inputs, targets)
+                    graphed = _GraphedStep(step_fn, inputs, targets, model)
                     logits, loss_ce, loss_kd, loss = graphed.out
                 except Exception as e:
                     print(f"[engine] step-graph capture failed "
