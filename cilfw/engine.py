@@ -170,7 +170,11 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         for inputs, targets, _tids in train_loader:
             inputs, targets = _to_device(inputs, targets, device, dtype)
             if can_graph and graphed is None and not first_of_epoch:
-                # capture on this batch; the capture run IS its training step
+                # capture on this batch; the capture run IS its training step.
+                # Drop the previous batch's autograd graph first: live loss/
+                # logits refs keep AccumulateGrad nodes pinned to the default
+                # stream, which breaks (segfaults) stream capture.
+                logits = loss_ce = loss_kd = loss = None  # noqa: F841
                 try:
                     graphed = _GraphedStep(step_fn, inputs, targets, model)
                     logits, loss_ce, loss_kd, loss = graphed.out
