@@ -114,15 +114,27 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   int4 areg[2 * NQ];
   __align__(16) bf16_t breg[NQ][8];
 
+  // incremental im2col decomposition (fast path): per q-chunk (r, s, c0)
+  // advance by compare/sub each K-step instead of div/mod every call
+  int inc_r[NQ], inc_s[NQ], inc_c0[NQ];
+  if (fast_a) {
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      int k = kt0 * BKT + q * 32 + ahalf * 16;
+      int rs = k / g.C;
+      inc_c0[q] = k - rs * g.C;
+      inc_r[q] = rs / g.S;
+      inc_s[q] = rs - inc_r[q] * g.S;
+    }
+  }
+
   auto stage_to_regs = [&](int kt) {
     const int k0 = kt * BKT;
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
-      // A chunk: 16 elems at col q*32 + ahalf*16 (within one (r,s): C%32==0)
+      // A chunk: 16 elems at col q*32 + ahalf*16 (within one (r,s): C%16==0)
       if (fast_a) {
-        int k = k0 + q * 32 + ahalf * 16;
-        int rs = k / g.C, c0 = k - rs * g.C;
-        int r = rs / g.S, s = rs - r * g.S;
+        int c0 = inc_c0[q], r = inc_r[q], s = inc_s[q];
         int hi = ahb + r, wi = awb + s;
         if (arow_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
           const int4* src = (const int4*)&x[(((long)an * g.H + hi) * g.W + wi)
@@ -133,6 +145,12 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
           areg[2 * q] = int4{0, 0, 0, 0};
           areg[2 * q + 1] = int4{0, 0, 0, 0};
         }
+        c0 += BKT;
+        while (c0 >= g.C) {
+          c0 -= g.C;
+          if (++s == g.S) { s = 0; ++r; }
+        }
+        inc_c0[q] = c0; inc_r[q] = r; inc_s[q] = s;
       } else {
         __align__(16) bf16_t tmp[16];
 #pragma unroll
@@ -271,14 +289,24 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
   int4 areg[2 * NQ];
   __align__(16) bf16_t breg[NQ][8];
 
+  int inc_r[NQ], inc_s[NQ], inc_kc[NQ];
+  if (fast_a) {
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      int k = kt0 * BKT + q * 32 + ahalf * 16;
+      int rs = k / g.K;
+      inc_kc[q] = k - rs * g.K;
+      inc_r[q] = rs / g.S;
+      inc_s[q] = rs - inc_r[q] * g.S;
+    }
+  }
+
   auto stage_to_regs = [&](int kt) {
     const int k0 = kt * BKT;
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
       if (fast_a) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
-        int k = k0 + q * 32 + ahalf * 16;
-        int rs = k / g.K, kc0 = k - rs * g.K;
-        int r = rs / g.S, s = rs - r * g.S;
+        int kc0 = inc_kc[q], r = inc_r[q], s = inc_s[q];
         int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
         bool ok = arow_ok && ho2 >= 0 && wo2 >= 0 &&
                   (ho2 % g.stride) == 0 && (wo2 % g.stride) == 0;
@@ -293,6 +321,12 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
           areg[2 * q] = int4{0, 0, 0, 0};
           areg[2 * q + 1] = int4{0, 0, 0, 0};
         }
+        kc0 += BKT;
+        while (kc0 >= g.K) {
+          kc0 -= g.K;
+          if (++s == g.S) { s = 0; ++r; }
+        }
+        inc_kc[q] = kc0; inc_r[q] = r; inc_s[q] = s;
       } else {
         __align__(16) bf16_t tmp[16];
 #pragma unroll
