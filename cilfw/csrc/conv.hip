@@ -23,7 +23,7 @@
 #define NTHREADS 256
 
 struct ConvGeom {
-  int N, H, W, C, K, R, S, stride, pad, Ho, Wo;
+  int N, H, W, C, K, R, S, stride, pad, pad2, Ho, Wo;  // pad2 = w-axis pad
 };
 
 // sum ksplit fp32 slabs [ns][len] -> bf16 out[len]
@@ -103,7 +103,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     awo = rem - aho * g.Wo;
   }
   const int ahb = aho * g.stride - g.pad;
-  const int awb = awo * g.stride - g.pad;
+  const int awb = awo * g.stride - g.pad2;
 
   f32x4 acc[4][2];
 #pragma unroll
@@ -307,7 +307,7 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     for (int q = 0; q < NQ; ++q) {
       if (fast_a) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
         int kc0 = inc_kc[q], r = inc_r[q], s = inc_s[q];
-        int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
+        int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
         bool ok = arow_ok && ho2 >= 0 && wo2 >= 0 &&
                   (ho2 % g.stride) == 0 && (wo2 % g.stride) == 0;
         int ho = ho2 / g.stride, wo = wo2 / g.stride;
@@ -336,7 +336,7 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
           if (arow_ok && k < RSK) {
             int rs = k / g.K, kc = k - rs * g.K;
             int r = rs / g.S, s = rs - r * g.S;
-            int ho2 = ahi + g.pad - r, wo2 = awi + g.pad - s;
+            int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
             if (ho2 >= 0 && wo2 >= 0 && (ho2 % g.stride) == 0 &&
                 (wo2 % g.stride) == 0) {
               int ho = ho2 / g.stride, wo = wo2 / g.stride;
@@ -484,7 +484,7 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     }
     if (fast_a) {
       int hi = ho * g.stride - g.pad + r_;
-      int wi = wo * g.stride - g.pad + s_;
+      int wi = wo * g.stride - g.pad2 + s_;
       if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
         *(int4*)adst = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
                                         * g.C + cbase_];
@@ -499,7 +499,7 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
           int rs = row / g.C, c = row - rs * g.C;
           int r = rs / g.S, s = rs - r * g.S;
           int hi = ho * g.stride - g.pad + r;
-          int wi = wo * g.stride - g.pad + s;
+          int wi = wo * g.stride - g.pad2 + s;
           if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
             v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c];
         }
@@ -593,6 +593,23 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
       }
 }
 
+// scatter a parity-class sub-grid result dxs (N,H2,W2,C) into
+// dx[:, ph::2, pw::2, :] (stride-2 bwd-data parity decomposition)
+__global__ __launch_bounds__(NTHREADS)
+void parity_scatter_kernel(const bf16_t* __restrict__ dxs,
+                           bf16_t* __restrict__ dx, int H, int W, int C,
+                           int ph, int pw, int H2, int W2, long total_sub) {
+  long i = (long)blockIdx.x * NTHREADS + threadIdx.x;
+  if (i >= total_sub) return;
+  int c = (int)(i % C);
+  long rest = i / C;
+  int w2 = (int)(rest % W2);
+  rest /= W2;
+  int h2 = (int)(rest % H2);
+  int n = (int)(rest / H2);
+  dx[(((long)n * H + ph + 2 * h2) * W + pw + 2 * w2) * C + c] = dxs[i];
+}
+
 // packed im2col pixel table: mt[m] = n<<20 | (ho*stride)<<10 | (wo*stride)
 __global__ __launch_bounds__(NTHREADS)
 void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
@@ -635,7 +652,7 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
                       int N, int H, int W, int C, int K, int R, int S,
                       int stride, int pad, int Ho, int Wo, int ksplit,
                       void* stream) {
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int fast_a = (C % 16 == 0);
@@ -675,7 +692,7 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
                            int N, int H, int W, int C, int K, int R, int S,
                            int stride, int pad, int Ho, int Wo, int ksplit,
                            void* stream) {
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);
@@ -702,6 +719,49 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   }
 }
 
+void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
+                               void* ws, int N, int H, int W, int C, int K,
+                               int R, int S, int padh, int padw, int Ho,
+                               int Wo, int ksplit, void* stream) {
+  // stride-1 sub-problem of the parity decomposition (asymmetric pads)
+  ConvGeom g{N, H, W, C, K, R, S, 1, padh, padw, Ho, Wo};
+  int M = N * H * W;
+  int RSK = R * S * K;
+  int fast_a = (K % 16 == 0);
+  int use64 = (RSK >= bk64_min_crs()) &&
+              (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
+  int nk = cdiv(RSK, use64 ? 64 : 32);
+  dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
+  if (use64)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
+                       nk, fast_a, ksplit);
+  else
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
+                       nk, fast_a, ksplit);
+  if (ksplit > 1) {
+    long len = (long)M * C;
+    hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                       dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                       dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                       (bf16_t*)dx, ksplit, len);
+  }
+}
+
+void cilfw_parity_scatter(const void* dxs, void* dx, int N, int H, int W,
+                          int C, int ph, int pw, int H2, int W2,
+                          void* stream) {
+  long total = (long)N * H2 * W2 * C;
+  hipLaunchKernelGGL(parity_scatter_kernel,
+                     dim3((int)cdiv((long)total, (long)NTHREADS)),
+                     dim3(NTHREADS), 0, (hipStream_t)stream,
+                     (const bf16_t*)dxs, (bf16_t*)dx, H, W, C, ph, pw, H2,
+                     W2, total);
+}
+
 int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
                                  int S) {
   int M = N * H * W;
@@ -715,7 +775,7 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
                              int K, int R, int S, int stride, int pad, int Ho,
                              int Wo, int nslices, void* stream) {
   (void)mt;  // kept in the ABI for the (cached) im2col table experiments
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int slice_len = cdiv(M, nslices);
