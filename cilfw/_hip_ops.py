@@ -139,39 +139,10 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
                                   c_i(Wo), _stream())
         _check("stride_scatter")
         return dx
-    if stride == 2 and (R > 1 or S > 1):
-        # parity decomposition: 4 stride-1 sub-convs over the (r,s) subsets
-        # that actually contribute to each (hi%2, wi%2) class — removes the
-        # 4x zero-structured MFMA work of the naive strided gather
-        dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
-        for ph in (0, 1):
-            H2 = (H - ph + 1) // 2
-            r0 = (ph + pad) % 2
-            nr = (R - r0 + 1) // 2
-            for pw in (0, 1):
-                W2 = (W_ - pw + 1) // 2
-                s0 = (pw + pad) % 2
-                ns = (S - s0 + 1) // 2
-                if H2 <= 0 or W2 <= 0:
-                    continue
-                assert nr > 0 and ns > 0, "empty parity class (unexpected R/S)"
-                wsub = w[r0::2, s0::2].contiguous()
-                dxs = torch.empty(N, H2, W2, C, dtype=torch.bfloat16,
-                                  device=dy.device)
-                ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H2, W2, C, K, nr,
-                                                       ns)
-                ws_ = (torch.empty(ks * N * H2 * W2 * C, dtype=torch.float32,
-                                   device=dy.device) if ks > 1 else None)
-                _lib.cilfw_conv2d_bwd_data_sub(
-                    _ptr(dy), _ptr(wsub), _ptr(dxs), _ptr(ws_), c_i(N),
-                    c_i(H2), c_i(W2), c_i(C), c_i(K), c_i(nr), c_i(ns),
-                    c_i((ph + pad - r0) // 2), c_i((pw + pad - s0) // 2),
-                    c_i(Ho), c_i(Wo), c_i(ks), _stream())
-                _lib.cilfw_parity_scatter(_ptr(dxs), _ptr(dx), c_i(N),
-                                          c_i(H), c_i(W_), c_i(C), c_i(ph),
-                                          c_i(pw), c_i(H2), c_i(W2),
-                                          _stream())
-        _check("conv2d_bwd_data_parity")
+    # NOTE: a 4-way parity decomposition of stride-2 bwd-data was measured
+    # SLOWER than the single zero-structured kernel (12 small launches vs one
+    # big one) — kernels kept (cilfw_conv2d_bwd_data_sub/parity_scatter) for a
+    # future fused single-launch variant.
         return dx
     dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
     ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H, W_, C, K, R, S)
