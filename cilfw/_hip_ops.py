@@ -143,7 +143,6 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
     # SLOWER than the single zero-structured kernel (12 small launches vs one
     # big one) — kernels kept (cilfw_conv2d_bwd_data_sub/parity_scatter) for a
     # future fused single-launch variant.
-        return dx
     dx = torch.empty(N, H, W_, C, dtype=torch.bfloat16, device=dy.device)
     ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H, W_, C, K, R, S)
     ws = (torch.empty(ks * N * H * W_ * C, dtype=torch.float32,
