@@ -21,6 +21,19 @@ import torch.nn.functional as F
 from ._backend import use_hip, ext
 
 
+_wgrad_stream = None
+
+
+def _wstream():
+    """Side stream for conv weight-grad kernels: dW is off the backward's
+    critical path (only dX feeds the next layer), so bwd-weight runs
+    concurrently with bwd-data — sum becomes max of the two."""
+    global _wgrad_stream
+    if _wgrad_stream is None:
+        _wgrad_stream = torch.cuda.Stream()
+    return _wgrad_stream
+
+
 def _to_nchw(x):
     return x.permute(0, 3, 1, 2).contiguous()
 
@@ -60,10 +73,16 @@ class Conv2dNHWC(torch.autograd.Function):
         dy = dy.contiguous()
         need_dx = ctx.needs_input_grad[0]  # stems skip the whole bwd-data pass
         if use_hip(dy):
+            cur = torch.cuda.current_stream()
+            ws = _wstream()
+            ws.wait_stream(cur)
+            with torch.cuda.stream(ws):  # dW concurrent with dX (fork/join)
+                dw = ext().conv2d_bwd_weight(dy, x, stride, padding,
+                                             wc.shape[0], wc.shape[1])
             dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1],
                                        x.shape[2]) if need_dx else None
-            dw = ext().conv2d_bwd_weight(dy, x, stride, padding,
-                                         wc.shape[0], wc.shape[1])
+            cur.wait_stream(ws)
+            dw.record_stream(cur)
             return dx, dw.to(ctx.w_dtype), None, None
         xf = _to_nchw(x).float()
         wf = wc.permute(3, 2, 0, 1).contiguous().float()
