@@ -73,7 +73,7 @@ void reduce_slabs_f32_kernel(const float* __restrict__ ws,
 
 // ============================== forward ==============================
 
-template <int BKT>
+template <int BKT, int BMX = BM>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        const bf16_t* __restrict__ w,
@@ -82,9 +82,11 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        int ksplit) {
   constexpr int NQ = BKT / 32;      // 16-elem chunks per thread in A staging
   constexpr int LPX = BKT + 8;
-  __shared__ bf16_t lds[2 * BM * LPX + 2 * BN * LPX];
-  const int AS0 = 0, BS0 = 2 * BM * LPX;
-  const int m0 = blockIdx.x * BM;
+  constexpr int NR = BMX / 128;     // A-tile rows per thread (1 or 2)
+  constexpr int MR = BMX / 32;      // M-fragments per wave (4 or 8)
+  __shared__ bf16_t lds[2 * BMX * LPX + 2 * BN * LPX];
+  const int AS0 = 0, BS0 = 2 * BMX * LPX;
+  const int m0 = blockIdx.x * BMX;
   const int ko0 = blockIdx.y * BN;
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
@@ -93,25 +95,31 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   const int kt1 = min(kt0 + steps, nk);
 
   const int arow = t >> 1, ahalf = t & 1;
-  int m = m0 + arow;
-  int an = 0, aho = 0, awo = 0;
-  bool arow_ok = m < M;
-  if (arow_ok) {
-    an = m / (g.Ho * g.Wo);
-    int rem = m - an * (g.Ho * g.Wo);
-    aho = rem / g.Wo;
-    awo = rem - aho * g.Wo;
-  }
-  const int ahb = aho * g.stride - g.pad;
-  const int awb = awo * g.stride - g.pad2;
-
-  f32x4 acc[4][2];
+  int an[NR], ahb[NR], awb[NR];
+  bool arow_ok[NR];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int rr = 0; rr < NR; ++rr) {
+    int m = m0 + arow + rr * 128;
+    arow_ok[rr] = m < M;
+    int n = 0, aho = 0, awo = 0;
+    if (arow_ok[rr]) {
+      n = m / (g.Ho * g.Wo);
+      int rem = m - n * (g.Ho * g.Wo);
+      aho = rem / g.Wo;
+      awo = rem - aho * g.Wo;
+    }
+    an[rr] = n;
+    ahb[rr] = aho * g.stride - g.pad;
+    awb[rr] = awo * g.stride - g.pad2;
+  }
+
+  f32x4 acc[MR][2];
+#pragma unroll
+  for (int i = 0; i < MR; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  int4 areg[2 * NQ];
+  int4 areg[NR][2 * NQ];
   __align__(16) bf16_t breg[NQ][8];
 
   // incremental im2col decomposition (fast path): per q-chunk (r, s, c0)
@@ -135,15 +143,18 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
       // A chunk: 16 elems at col q*32 + ahalf*16 (within one (r,s): C%16==0)
       if (fast_a) {
         int c0 = inc_c0[q], r = inc_r[q], s = inc_s[q];
-        int hi = ahb + r, wi = awb + s;
-        if (arow_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
-          const int4* src = (const int4*)&x[(((long)an * g.H + hi) * g.W + wi)
-                                            * g.C + c0];
-          areg[2 * q] = src[0];
-          areg[2 * q + 1] = src[1];
-        } else {
-          areg[2 * q] = int4{0, 0, 0, 0};
-          areg[2 * q + 1] = int4{0, 0, 0, 0};
+#pragma unroll
+        for (int rr = 0; rr < NR; ++rr) {
+          int hi = ahb[rr] + r, wi = awb[rr] + s;
+          if (arow_ok[rr] && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
+            const int4* src = (const int4*)&x[(((long)an[rr] * g.H + hi)
+                                               * g.W + wi) * g.C + c0];
+            areg[rr][2 * q] = src[0];
+            areg[rr][2 * q + 1] = src[1];
+          } else {
+            areg[rr][2 * q] = int4{0, 0, 0, 0};
+            areg[rr][2 * q + 1] = int4{0, 0, 0, 0};
+          }
         }
         c0 += BKT;
         while (c0 >= g.C) {
@@ -152,22 +163,25 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         }
         inc_c0[q] = c0; inc_r[q] = r; inc_s[q] = s;
       } else {
-        __align__(16) bf16_t tmp[16];
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          int k = k0 + q * 32 + ahalf * 16 + j;
-          bf16_t v = 0;
-          if (arow_ok && k < CRS) {
-            int rs = k / g.C, c = k - rs * g.C;
-            int r = rs / g.S, s = rs - r * g.S;
-            int hi = ahb + r, wi = awb + s;
-            if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-              v = x[(((long)an * g.H + hi) * g.W + wi) * g.C + c];
+        for (int rr = 0; rr < NR; ++rr) {
+          __align__(16) bf16_t tmp[16];
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            int k = k0 + q * 32 + ahalf * 16 + j;
+            bf16_t v = 0;
+            if (arow_ok[rr] && k < CRS) {
+              int rs = k / g.C, c = k - rs * g.C;
+              int r = rs / g.S, s = rs - r * g.S;
+              int hi = ahb[rr] + r, wi = awb[rr] + s;
+              if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+                v = x[(((long)an[rr] * g.H + hi) * g.W + wi) * g.C + c];
+            }
+            tmp[j] = v;
           }
-          tmp[j] = v;
+          areg[rr][2 * q] = *(int4*)&tmp[0];
+          areg[rr][2 * q + 1] = *(int4*)&tmp[8];
         }
-        areg[2 * q] = *(int4*)&tmp[0];
-        areg[2 * q + 1] = *(int4*)&tmp[8];
       }
       // B chunk: thread owns one Bs row (cout n = t&63) and 8 k's
       // (kgrp = t>>6): strided reads are coalesced ACROSS lanes (n contiguous)
@@ -183,12 +197,17 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   };
 
   auto regs_to_lds = [&](int buf) {
-    bf16_t* As = &lds[AS0 + buf * BM * LPX];
+    bf16_t* As = &lds[AS0 + buf * BMX * LPX];
     bf16_t* Bs = &lds[BS0 + buf * BN * LPX];
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
-      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16] = areg[2 * q];
-      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16 + 8] = areg[2 * q + 1];
+#pragma unroll
+      for (int rr = 0; rr < NR; ++rr) {
+        *(int4*)&As[(arow + rr * 128) * LPX + q * 32 + ahalf * 16] =
+            areg[rr][2 * q];
+        *(int4*)&As[(arow + rr * 128) * LPX + q * 32 + ahalf * 16 + 8] =
+            areg[rr][2 * q + 1];
+      }
       const int bn = t & 63, bk8 = (t >> 6) * 8 + q * 32;
       *(int4*)&Bs[bn * LPX + bk8] = *(int4*)breg[q];
     }
@@ -204,14 +223,14 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     int cur = (kt - kt0) & 1;
     if (kt + 1 < kt1) stage_to_regs(kt + 1);
     {
-      const bf16_t* As = &lds[AS0 + cur * BM * LPX];
+      const bf16_t* As = &lds[AS0 + cur * BMX * LPX];
       const bf16_t* Bs = &lds[BS0 + cur * BN * LPX];
 #pragma unroll
       for (int q = 0; q < NQ; ++q) {
         const int kb = fi.quad * 8 + q * 32;
 #pragma unroll
-        for (int mr = 0; mr < 4; ++mr) {
-          int row = wr * 64 + mr * 16 + fi.half;
+        for (int mr = 0; mr < MR; ++mr) {
+          int row = wr * (BMX / 2) + mr * 16 + fi.half;
           bf16x8 a = *(const bf16x8*)&As[row * LPX + kb];
 #pragma unroll
           for (int nr = 0; nr < 2; ++nr) {
@@ -231,12 +250,12 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   }
 
 #pragma unroll
-  for (int mr = 0; mr < 4; ++mr)
+  for (int mr = 0; mr < MR; ++mr)
 #pragma unroll
     for (int nr = 0; nr < 2; ++nr)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = m0 + wr * 64 + mr * 16 + fi.quad * 4 + r;
+        int row = m0 + wr * (BMX / 2) + mr * 16 + fi.quad * 4 + r;
         int col = ko0 + wc * 32 + nr * 16 + fi.half;
         if (row < M && col < g.K) {
           if (ksplit > 1)
@@ -648,6 +667,16 @@ static int pick_ksplit(int nblocks, int nk) {
 
 extern "C" {
 
+static int fwd_bm256_min_m() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_BM256_MIN");
+    v = e ? atoi(e) : 65536;
+    if (v <= 0) v = 1 << 30;
+  }
+  return v;
+}
+
 void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
                       int N, int H, int W, int C, int K, int R, int S,
                       int stride, int pad, int Ho, int Wo, int ksplit,
@@ -656,12 +685,19 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int fast_a = (C % 16 == 0);
-  // BKT=64 only when the grid is too small to exploit BKT=32's 2x occupancy
-  int use64 = (CRS >= bk64_min_crs()) &&
+  // BM=256 doubles MFMA-per-barrier when M is large enough to keep the grid
+  // full; BKT=64 only when the grid is too small for BKT=32's 2x occupancy
+  int bm = (M >= fwd_bm256_min_m()) ? 256 : BM;
+  int use64 = bm == BM && (CRS >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(K, BN) * ksplit < 768);
   int nk = cdiv(CRS, use64 ? 64 : 32);
-  dim3 grid(cdiv(M, BM), cdiv(K, BN), ksplit);
-  if (use64)
+  dim3 grid(cdiv(M, bm), cdiv(K, BN), ksplit);
+  if (bm == 256)
+    hipLaunchKernelGGL((conv2d_fwd_kernel<32, 256>), grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)x,
+                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
+                       nk, fast_a, ksplit);
+  else if (use64)
     hipLaunchKernelGGL((conv2d_fwd_kernel<64>), grid, dim3(NTHREADS), 0,
                        (hipStream_t)stream, (const bf16_t*)x,
                        (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
