@@ -671,8 +671,8 @@ static int fwd_bm256_min_m() {
   static int v = -1;
   if (v < 0) {
     const char* e = getenv("CILFW_CONV_BM256_MIN");
-    v = e ? atoi(e) : 65536;
-    if (v <= 0) v = 1 << 30;
+    v = e ? atoi(e) : 0;  // measured: BM=256 loses to BM=128's occupancy at
+    if (v <= 0) v = 1 << 30;  // every ResNet shape — off by default
   }
   return v;
 }
