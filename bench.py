@@ -91,15 +91,15 @@ def main():
             with torch.cuda.stream(tstream), torch.no_grad():
                 t_logits, _ = teacher(x)
         logits, _ = model(x)
-        loss = ops.cross_entropy(logits.float(), y)
         if teacher is not None:
             if tstream is not None:
                 torch.cuda.current_stream().wait_stream(tstream)
             else:
                 with torch.no_grad():
                     t_logits, _ = teacher(x)
-            loss = loss + 0.5 * ops.kd_loss(logits[:, :known].float(),
-                                            t_logits.float(), 2.0)
+            loss, _lce, _lkd = ops.wa_loss(logits, t_logits, y, 0.0, 2.0, 0.5)
+        else:
+            loss, _lce, _lkd = ops.wa_loss(logits, None, y, 0.0, 2.0, 0.5)
         loss.backward()
         engine.finalize()
         opt.step()

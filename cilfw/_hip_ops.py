@@ -59,6 +59,8 @@ _PROTOS = {
     "cilfw_ce_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_fwd": [c_vp] * 6 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
+    "cilfw_wa_loss_fwd": [c_vp] * 8 + [c_i] * 3 + [c_f] * 3 + [c_vp],
+    "cilfw_wa_loss_bwd": [c_vp] * 6 + [c_i] * 3 + [c_f] * 3 + [c_vp],
     "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_f, c_f, c_f, c_vp],
     "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_herding_select": [c_vp] * 3 + [c_i] * 3 + [c_vp],
@@ -405,6 +407,41 @@ def kd_bwd(ps, pt, T, dloss):
                       c_f(T), _stream())
     _check("kd_bwd")
     return ds
+
+
+def wa_loss_fwd(s_logits, t_logits, targets, smooth, T, lam):
+    """Fused CE(+smoothing) + lambda*KD over bf16 logits. t_logits may be
+    None (plain CE). Returns (loss_total, loss_ce, loss_kd, probs, ps, pt)."""
+    _bf16(s_logits, "wa_loss.s")
+    M, C = s_logits.shape
+    Ck = 0
+    if t_logits is not None:
+        _bf16(t_logits, "wa_loss.t")
+        Ck = t_logits.shape[1]
+    dev = s_logits.device
+    probs = torch.empty(M, C, dtype=torch.float32, device=dev)
+    ps = torch.empty(M, max(Ck, 1), dtype=torch.float32, device=dev)
+    pt = torch.empty(M, max(Ck, 1), dtype=torch.float32, device=dev)
+    rl2 = torch.empty(2 * M, dtype=torch.float32, device=dev)
+    out3 = torch.empty(3, dtype=torch.float32, device=dev)
+    _lib.cilfw_wa_loss_fwd(_ptr(s_logits), _ptr(t_logits),
+                           _ptr(targets.contiguous()), _ptr(probs), _ptr(ps),
+                           _ptr(pt), _ptr(rl2), _ptr(out3), c_i(M), c_i(C),
+                           c_i(Ck), c_f(smooth), c_f(T), c_f(lam), _stream())
+    _check("wa_loss_fwd")
+    return out3[2], out3[0], out3[1], probs, ps, pt
+
+
+def wa_loss_bwd(probs, ps, pt, targets, dtotal, smooth, T, lam, Ck):
+    M, C = probs.shape
+    dlogits = torch.empty(M, C, dtype=torch.bfloat16, device=probs.device)
+    dl = dtotal.float().reshape(1).contiguous()
+    _lib.cilfw_wa_loss_bwd(_ptr(probs), _ptr(ps), _ptr(pt),
+                           _ptr(targets.contiguous()), _ptr(dl),
+                           _ptr(dlogits), c_i(M), c_i(C), c_i(Ck),
+                           c_f(smooth), c_f(T), c_f(lam), _stream())
+    _check("wa_loss_bwd")
+    return dlogits
 
 
 # ------------------------------------------------------------- optimizer / misc

@@ -149,6 +149,155 @@ void kd_bwd_kernel(const float* __restrict__ ps, const float* __restrict__ pt,
   ds[i] = (ps[i] - pt[i]) * (T / M) * dloss[0];
 }
 
+// ------------------------------------------- fully fused WA loss (CE + KD)
+// One kernel pair for the whole WA training objective: CE(+label smoothing)
+// over all C classes + lambda * SoftTarget KD over the first Ck (teacher)
+// classes, reading the bf16 logits DIRECTLY (no fp32 cast round-trips) and
+// emitting bf16 dlogits for the classifier backward. One block per row.
+
+__global__ __launch_bounds__(NT)
+void wa_loss_fwd_kernel(const bf16_t* __restrict__ slog,
+                        const bf16_t* __restrict__ tlog,
+                        const long* __restrict__ targets,
+                        float* __restrict__ probs,   // (M, C)
+                        float* __restrict__ ps,      // (M, Ck)
+                        float* __restrict__ pt,      // (M, Ck)
+                        float* __restrict__ rl_ce, float* __restrict__ rl_kd,
+                        int M, int C, int Ck, float smooth, float T) {
+  const int row = blockIdx.x;
+  const bf16_t* sr = slog + (long)row * C;
+  __shared__ float red[NT / WAVE];
+
+  // ---- CE over all C ----
+  float mx = -3.4e38f;
+  for (int c = threadIdx.x; c < C; c += NT) mx = fmaxf(mx, bf2f(sr[c]));
+  for (int o = 32; o > 0; o >>= 1) mx = fmaxf(mx, __shfl_xor(mx, o));
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = mx;
+  __syncthreads();
+  mx = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+  float se = 0.f;
+  for (int c = threadIdx.x; c < C; c += NT) se += __expf(bf2f(sr[c]) - mx);
+  for (int o = 32; o > 0; o >>= 1) se += __shfl_xor(se, o);
+  __syncthreads();
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = se;
+  __syncthreads();
+  se = red[0] + red[1] + red[2] + red[3];
+  const float lse = __logf(se) + mx;
+  float sum_logp = 0.f;
+  for (int c = threadIdx.x; c < C; c += NT) {
+    float logp = bf2f(sr[c]) - lse;
+    probs[(long)row * C + c] = __expf(logp);
+    sum_logp += logp;
+  }
+  if (smooth > 0.f) {
+    for (int o = 32; o > 0; o >>= 1) sum_logp += __shfl_xor(sum_logp, o);
+    __syncthreads();
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = sum_logp;
+    __syncthreads();
+    sum_logp = red[0] + red[1] + red[2] + red[3];
+  }
+  if (threadIdx.x == 0) {
+    float nll = lse - bf2f(sr[targets[row]]);
+    rl_ce[row] = (1.f - smooth) * nll - smooth * (sum_logp / C);
+  }
+
+  if (Ck <= 0) {
+    if (threadIdx.x == 0) rl_kd[row] = 0.f;
+    return;
+  }
+
+  // ---- KD over the first Ck columns ----
+  const bf16_t* tr = tlog + (long)row * Ck;
+  float lsek[2];
+#pragma unroll
+  for (int which = 0; which < 2; ++which) {
+    const bf16_t* lr = which == 0 ? sr : tr;
+    float m2 = -3.4e38f;
+    for (int c = threadIdx.x; c < Ck; c += NT)
+      m2 = fmaxf(m2, bf2f(lr[c]) / T);
+    for (int o = 32; o > 0; o >>= 1) m2 = fmaxf(m2, __shfl_xor(m2, o));
+    __syncthreads();
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = m2;
+    __syncthreads();
+    m2 = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    float s2 = 0.f;
+    for (int c = threadIdx.x; c < Ck; c += NT)
+      s2 += __expf(bf2f(lr[c]) / T - m2);
+    for (int o = 32; o > 0; o >>= 1) s2 += __shfl_xor(s2, o);
+    __syncthreads();
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = s2;
+    __syncthreads();
+    s2 = red[0] + red[1] + red[2] + red[3];
+    lsek[which] = __logf(s2) + m2;
+    float* out = which == 0 ? ps : pt;
+    for (int c = threadIdx.x; c < Ck; c += NT)
+      out[(long)row * Ck + c] = __expf(bf2f(lr[c]) / T - lsek[which]);
+  }
+  float l = 0.f;
+  for (int c = threadIdx.x; c < Ck; c += NT) {
+    float logps = bf2f(sr[c]) / T - lsek[0];
+    float logpt = bf2f(tr[c]) / T - lsek[1];
+    l += __expf(logpt) * (logpt - logps);
+  }
+  for (int o = 32; o > 0; o >>= 1) l += __shfl_xor(l, o);
+  __syncthreads();
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = l;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    l = red[0] + red[1] + red[2] + red[3];
+    rl_kd[row] = l * T * T;
+  }
+}
+
+// out[0]=mean(ce), out[1]=mean(kd), out[2]=mean(ce)+lam*mean(kd)
+__global__ __launch_bounds__(NT)
+void loss_mean3_kernel(const float* __restrict__ rl_ce,
+                       const float* __restrict__ rl_kd,
+                       float* __restrict__ out, int M, float lam) {
+  __shared__ float red[2][NT];
+  float a = 0.f, b = 0.f;
+  for (int i = threadIdx.x; i < M; i += NT) {
+    a += rl_ce[i];
+    b += rl_kd[i];
+  }
+  red[0][threadIdx.x] = a;
+  red[1][threadIdx.x] = b;
+  __syncthreads();
+  for (int o = NT / 2; o > 0; o >>= 1) {
+    if (threadIdx.x < o) {
+      red[0][threadIdx.x] += red[0][threadIdx.x + o];
+      red[1][threadIdx.x] += red[1][threadIdx.x + o];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    out[0] = red[0][0] / M;
+    out[1] = red[1][0] / M;
+    out[2] = out[0] + lam * out[1];
+  }
+}
+
+// dlogits(bf16) = dtotal * [ dCE + lam*dKD ]   (dtotal read from device)
+__global__ __launch_bounds__(NT)
+void wa_loss_bwd_kernel(const float* __restrict__ probs,
+                        const float* __restrict__ ps,
+                        const float* __restrict__ pt,
+                        const long* __restrict__ targets,
+                        const float* __restrict__ dtotal,
+                        bf16_t* __restrict__ dlogits, int M, int C, int Ck,
+                        float smooth, float T, float lam) {
+  long i = (long)blockIdx.x * NT + threadIdx.x;
+  if (i >= (long)M * C) return;
+  int row = (int)(i / C), c = (int)(i % C);
+  float g = probs[i] - smooth / C;
+  if (c == (int)targets[row]) g -= (1.f - smooth);
+  if (c < Ck) {
+    long j = (long)row * Ck + c;
+    g += lam * T * (ps[j] - pt[j]);
+  }
+  dlogits[i] = f2bf(g * dtotal[0] / M);
+}
+
 // ------------------------------------------------------------------- fused SGD
 // g = grad + wd*p; m = mu*m + g; p -= lr*m   — one kernel over the flat buffers.
 
@@ -368,6 +517,34 @@ void cilfw_herding_select(const void* f, const void* mu, void* order, int n,
   hipLaunchKernelGGL(herding_kernel, dim3(1), dim3(NT), shmem,
                      (hipStream_t)stream, (const float*)f, (const float*)mu,
                      (long*)order, n, D, m);
+}
+
+void cilfw_wa_loss_fwd(const void* slog, const void* tlog,
+                       const void* targets, void* probs, void* ps, void* pt,
+                       void* rowloss2, void* out3, int M, int C, int Ck,
+                       float smooth, float T, float lam, void* stream) {
+  float* rl_ce = (float*)rowloss2;
+  float* rl_kd = rl_ce + M;
+  hipLaunchKernelGGL(wa_loss_fwd_kernel, dim3(M), dim3(NT), 0,
+                     (hipStream_t)stream, (const bf16_t*)slog,
+                     (const bf16_t*)tlog, (const long*)targets, (float*)probs,
+                     (float*)ps, (float*)pt, rl_ce, rl_kd, M, C, Ck, smooth,
+                     T);
+  hipLaunchKernelGGL(loss_mean3_kernel, dim3(1), dim3(NT), 0,
+                     (hipStream_t)stream, rl_ce, rl_kd, (float*)out3, M, lam);
+}
+
+void cilfw_wa_loss_bwd(const void* probs, const void* ps, const void* pt,
+                       const void* targets, const void* dtotal, void* dlogits,
+                       int M, int C, int Ck, float smooth, float T, float lam,
+                       void* stream) {
+  long total = (long)M * C;
+  hipLaunchKernelGGL(wa_loss_bwd_kernel,
+                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                     (hipStream_t)stream, (const float*)probs,
+                     (const float*)ps, (const float*)pt,
+                     (const long*)targets, (const float*)dtotal,
+                     (bf16_t*)dlogits, M, C, Ck, smooth, T, lam);
 }
 
 }  // extern "C"

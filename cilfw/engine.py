@@ -60,7 +60,7 @@ def evaluate(model, loader, device, args, header="Test:"):
     for inputs, targets, _tids in loader:
         inputs, targets = _to_device(inputs, targets, device, dtype)
         logits, _ = model(inputs)
-        loss = ops.cross_entropy(logits.float(), targets)
+        loss, _ce, _kd = ops.wa_loss(logits, None, targets)
         accs = ops.accuracy(logits, targets, topk=topk)
         bs = targets.shape[0]
         metric_logger.update(loss=loss.item())
@@ -131,19 +131,19 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
             with torch.cuda.stream(tstream), torch.no_grad():
                 t_logits, _ = teacher(inputs)
         logits, _features = model(inputs)
-        loss_ce = ops.cross_entropy(logits.float(), targets, args.smooth)
         if teacher is not None:
             if tstream is not None:
                 torch.cuda.current_stream().wait_stream(tstream)
             else:
                 with torch.no_grad():
                     t_logits, _ = teacher(inputs)
-            loss_kd = ops.kd_loss(logits[:, :known].float(),
-                                  t_logits.float(), args.kd_temperature)
-            loss = loss_ce + lambda_kd * loss_kd
+            loss, loss_ce, loss_kd = ops.wa_loss(
+                logits, t_logits, targets, args.smooth,
+                args.kd_temperature, lambda_kd)
         else:
-            loss_kd = torch.zeros((), device=logits.device)
-            loss = loss_ce
+            loss, loss_ce, loss_kd = ops.wa_loss(
+                logits, None, targets, args.smooth,
+                args.kd_temperature, lambda_kd)
         loss.backward()
         engine.finalize()
         if update:

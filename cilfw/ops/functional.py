@@ -505,6 +505,67 @@ def max_pool(x, kernel=3, stride=2, padding=1):
     return MaxPoolNHWC.apply(x, kernel, stride, padding)
 
 
+# ----------------------------------------------------- fused WA loss (CE + KD)
+
+
+class WALoss(torch.autograd.Function):
+    """The whole WA training objective in one fused kernel pair:
+    loss = CE(logits, targets; smoothing) + lam * SoftTarget(logits[:, :Ck],
+    t_logits; T). Reads bf16 logits directly and emits bf16 dlogits (no fp32
+    cast round-trips). Only the TOTAL loss output participates in autograd;
+    loss_ce / loss_kd are detached (for logging)."""
+
+    @staticmethod
+    def forward(ctx, logits, t_logits, targets, smooth, T, lam):
+        ctx.params = (smooth, T, lam,
+                      0 if t_logits is None else t_logits.shape[1])
+        if use_hip(logits):
+            total, ce, kd, probs, ps, pt = ext().wa_loss_fwd(
+                logits, t_logits, targets, smooth, T, lam)
+            ctx.save_for_backward(probs, ps, pt, targets)
+            return total, ce, kd
+        lf = logits.float()
+        logp = F.log_softmax(lf, dim=1)
+        nll = -logp.gather(1, targets.view(-1, 1)).squeeze(1)
+        ce = ((1 - smooth) * nll - smooth * logp.mean(dim=1)).mean() \
+            if smooth > 0 else nll.mean()
+        probs = logp.exp()
+        if t_logits is not None:
+            Ck = t_logits.shape[1]
+            sf = lf[:, :Ck] / T
+            tf = t_logits.float() / T
+            ps, pt = F.softmax(sf, 1), F.softmax(tf, 1)
+            kd = (pt * (F.log_softmax(tf, 1) - F.log_softmax(sf, 1))
+                  ).sum() / lf.shape[0] * (T * T)
+        else:
+            ps = pt = lf.new_zeros(lf.shape[0], 1)
+            kd = lf.new_zeros(())
+        ctx.save_for_backward(probs, ps, pt, targets)
+        return ce + lam * kd, ce, kd
+
+    @staticmethod
+    def backward(ctx, dtotal, _dce, _dkd):
+        # only the TOTAL output participates in training backward; the ce/kd
+        # outputs are logging-only (their grads, if any, are ignored)
+        probs, ps, pt, targets = ctx.saved_tensors
+        smooth, T, lam, Ck = ctx.params
+        if use_hip(probs):
+            dlogits = ext().wa_loss_bwd(probs, ps, pt, targets, dtotal,
+                                        smooth, T, lam, Ck)
+            return dlogits, None, None, None, None, None
+        M, C = probs.shape
+        g = probs - smooth / C
+        g.scatter_add_(1, targets.view(-1, 1),
+                       torch.full((M, 1), -(1 - smooth), dtype=g.dtype))
+        if Ck > 0:
+            g[:, :Ck] += lam * T * (ps - pt)
+        return (g * (dtotal / M), None, None, None, None, None)
+
+
+def wa_loss(logits, t_logits, targets, smooth=0.0, T=2.0, lam=0.5):
+    return WALoss.apply(logits, t_logits, targets, smooth, T, lam)
+
+
 # ------------------------------------------------------------------ top-k accuracy
 
 

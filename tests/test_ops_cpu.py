@@ -204,3 +204,36 @@ def test_bn_add_relu_fused_matches_composite():
     y2.backward(dy)
     for a, c in [(x, x2), (res, res2), (g, g2), (b, b2)]:
         assert torch.allclose(a.grad, c.grad, atol=1e-5)
+
+
+def test_wa_loss_matches_composite():
+    torch.manual_seed(4)
+    M, C, Ck = 9, 20, 12
+    s = torch.randn(M, C, requires_grad=True)
+    t = torch.randn(M, Ck)
+    y = torch.randint(0, C, (M,))
+    lam, T, sm = 0.5, 2.0, 0.1
+    total, ce, kd = ops.wa_loss(s, t, y, sm, T, lam)
+    s2 = s.detach().requires_grad_()
+    ce2 = F.cross_entropy(s2, y, label_smoothing=sm)
+    kd2 = F.kl_div(F.log_softmax(s2[:, :Ck] / T, 1),
+                   F.softmax(t / T, 1), reduction="batchmean") * T * T
+    tot2 = ce2 + lam * kd2
+    assert torch.allclose(total, tot2, atol=1e-6)
+    assert torch.allclose(ce, ce2, atol=1e-6)
+    assert torch.allclose(kd, kd2, atol=1e-6)
+    total.backward()
+    tot2.backward()
+    assert torch.allclose(s.grad, s2.grad, atol=1e-6)
+
+
+def test_wa_loss_no_teacher():
+    torch.manual_seed(5)
+    s = torch.randn(7, 11, requires_grad=True)
+    y = torch.randint(0, 11, (7,))
+    total, ce, kd = ops.wa_loss(s, None, y, 0.0, 2.0, 0.5)
+    assert kd.item() == 0.0
+    ref = F.cross_entropy(s.detach(), y)
+    assert torch.allclose(total, ref, atol=1e-6)
+    total.backward()
+    assert torch.isfinite(s.grad).all()

@@ -304,3 +304,20 @@ def test_bn_add_relu_gpu():
     _cmp(xg.grad, xc.grad, rtol=0.03, atol=0.03, what="bn_add_relu dx")
     _cmp(rg.grad, rc.grad, what="bn_add_relu dres")
     _cmp(gg.grad, gc.grad, what="bn_add_relu dgamma")
+
+
+def test_wa_loss_gpu():
+    torch.manual_seed(40)
+    M, C, Ck = 64, 110, 90
+    s = torch.randn(M, C).to(torch.bfloat16)
+    t = torch.randn(M, Ck).to(torch.bfloat16)
+    y = torch.randint(0, C, (M,))
+    sg = s.cuda().requires_grad_()
+    sc = s.clone().requires_grad_()
+    outg = CF.wa_loss(sg, t.cuda(), y.cuda(), 0.1, 2.0, 0.5)
+    outc = CF.wa_loss(sc, t, y, 0.1, 2.0, 0.5)
+    for a, b, w in zip(outg, outc, ("total", "ce", "kd")):
+        _cmp(a, b, rtol=1e-3, atol=1e-3, what=f"wa_loss {w}")
+    outg[0].backward()
+    outc[0].backward()
+    _cmp(sg.grad, sc.grad, rtol=2e-2, atol=2e-3, what="wa_loss grad")
