@@ -242,7 +242,10 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         }
       }
     }
-    __syncthreads();
+    // single barrier per K-step: writes target buf[cur^1], whose readers all
+    // finished before the PREVIOUS barrier — the post-MFMA barrier is
+    // unnecessary (fast waves cannot overwrite buf[cur]: the next write to it
+    // happens after this barrier, which slow readers must reach first)
     if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
       __syncthreads();
@@ -422,7 +425,10 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
         }
       }
     }
-    __syncthreads();
+    // single barrier per K-step: writes target buf[cur^1], whose readers all
+    // finished before the PREVIOUS barrier — the post-MFMA barrier is
+    // unnecessary (fast waves cannot overwrite buf[cur]: the next write to it
+    // happens after this barrier, which slow readers must reach first)
     if (kt + 1 < kt1) {
       regs_to_lds(cur ^ 1);
       __syncthreads();
@@ -590,8 +596,7 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
         }
       }
     }
-    __syncthreads();
-    if (kt + 1 < nk) {
+    if (kt + 1 < nk) {  // single barrier per K-step (see fwd kernel note)
       regs_to_lds(cur ^ 1);
       __syncthreads();
     }
