@@ -73,7 +73,7 @@ void reduce_slabs_f32_kernel(const float* __restrict__ ws,
 
 // ============================== forward ==============================
 
-template <int BKT, int BMX = BM>
+template <int BKT, int BMX = BM, int BNX = BN>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        const bf16_t* __restrict__ w,
@@ -84,10 +84,11 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   constexpr int LPX = BKT + 8;
   constexpr int NR = BMX / 128;     // A-tile rows per thread (1 or 2)
   constexpr int MR = BMX / 32;      // M-fragments per wave (4 or 8)
-  __shared__ bf16_t lds[2 * BMX * LPX + 2 * BN * LPX];
+  constexpr int NRC = BNX / 32;     // N-fragments per wave (2 or 4)
+  __shared__ bf16_t lds[2 * BMX * LPX + 2 * BNX * LPX];
   const int AS0 = 0, BS0 = 2 * BMX * LPX;
   const int m0 = blockIdx.x * BMX;
-  const int ko0 = blockIdx.y * BN;
+  const int ko0 = blockIdx.y * BNX;
   const int t = threadIdx.x;
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
   const int steps = (nk + ksplit - 1) / ksplit;
@@ -113,14 +114,14 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
     awb[rr] = awo * g.stride - g.pad2;
   }
 
-  f32x4 acc[MR][2];
+  f32x4 acc[MR][NRC];
 #pragma unroll
   for (int i = 0; i < MR; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
+    for (int j = 0; j < NRC; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
   int4 areg[NR][2 * NQ];
-  __align__(16) bf16_t breg[NQ][8];
+  __align__(16) bf16_t breg[NQ][BNX / 64][8];
 
   // incremental im2col decomposition (fast path): per q-chunk (r, s, c0)
   // advance by compare/sub each K-step instead of div/mod every call
@@ -183,15 +184,19 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
           areg[rr][2 * q + 1] = *(int4*)&tmp[8];
         }
       }
-      // B chunk: thread owns one Bs row (cout n = t&63) and 8 k's
-      // (kgrp = t>>6): strided reads are coalesced ACROSS lanes (n contiguous)
-      // and the LDS write is one b128 (conflict-free 8-lane groups).
-      const int bn = t & 63, bk8 = (t >> 6) * 8 + q * 32;
-      const bool n_ok = ko0 + bn < g.K;
+      // B chunk: thread owns one Bs row (cout n) per 64-col group and 8 k's:
+      // strided reads coalesce ACROSS lanes (n contiguous); each LDS write is
+      // one b128 (conflict-free 8-lane groups).
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int k = k0 + bk8 + j;
-        breg[q][j] = (n_ok && k < CRS) ? w[(long)k * g.K + ko0 + bn] : 0;
+      for (int h = 0; h < BNX / 64; ++h) {
+        const int bn = (t & 63) + h * 64;
+        const int bk8 = ((t >> 6) & 3) * 8 + q * 32;
+        const bool n_ok = ko0 + bn < g.K;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int k = k0 + bk8 + j;
+          breg[q][h][j] = (n_ok && k < CRS) ? w[(long)k * g.K + ko0 + bn] : 0;
+        }
       }
     }
   };
@@ -208,8 +213,12 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         *(int4*)&As[(arow + rr * 128) * LPX + q * 32 + ahalf * 16 + 8] =
             areg[rr][2 * q + 1];
       }
-      const int bn = t & 63, bk8 = (t >> 6) * 8 + q * 32;
-      *(int4*)&Bs[bn * LPX + bk8] = *(int4*)breg[q];
+#pragma unroll
+      for (int h = 0; h < BNX / 64; ++h) {
+        const int bn = (t & 63) + h * 64;
+        const int bk8 = ((t >> 6) & 3) * 8 + q * 32;
+        *(int4*)&Bs[bn * LPX + bk8] = *(int4*)breg[q][h];
+      }
     }
   };
 
@@ -233,8 +242,8 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
           int row = wr * (BMX / 2) + mr * 16 + fi.half;
           bf16x8 a = *(const bf16x8*)&As[row * LPX + kb];
 #pragma unroll
-          for (int nr = 0; nr < 2; ++nr) {
-            int col = wc * 32 + nr * 16 + fi.half;
+          for (int nr = 0; nr < NRC; ++nr) {
+            int col = wc * (BNX / 2) + nr * 16 + fi.half;
             bf16x8 b = *(const bf16x8*)&Bs[col * LPX + kb];
             acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a, b, acc[mr][nr], 0, 0, 0);
@@ -255,11 +264,11 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
   for (int mr = 0; mr < MR; ++mr)
 #pragma unroll
-    for (int nr = 0; nr < 2; ++nr)
+    for (int nr = 0; nr < NRC; ++nr)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * (BMX / 2) + mr * 16 + fi.quad * 4 + r;
-        int col = ko0 + wc * 32 + nr * 16 + fi.half;
+        int col = ko0 + wc * (BNX / 2) + nr * 16 + fi.half;
         if (row < M && col < g.K) {
           if (ksplit > 1)
             ws[((long)blockIdx.z * M + row) * g.K + col] = acc[mr][nr][r];
@@ -672,6 +681,16 @@ static int pick_ksplit(int nblocks, int nk) {
 
 extern "C" {
 
+static int fwd_bn128_min_wgs() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_BN128_MIN_WGS");
+    v = e ? atoi(e) : 256;
+    if (v <= 0) v = 1 << 30;
+  }
+  return v;
+}
+
 static int fwd_bm256_min_m() {
   static int v = -1;
   if (v < 0) {
@@ -693,11 +712,19 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   // BM=256 doubles MFMA-per-barrier when M is large enough to keep the grid
   // full; BKT=64 only when the grid is too small for BKT=32's 2x occupancy
   int bm = (M >= fwd_bm256_min_m()) ? 256 : BM;
-  int use64 = bm == BM && (CRS >= bk64_min_crs()) &&
+  // BN=128 doubles A-tile reuse when K is wide and the grid stays full
+  int bn128 = (bm == BM) && (K % 128 == 0) &&
+              (cdiv(M, BM) * cdiv(K, 128) * ksplit >= fwd_bn128_min_wgs());
+  int use64 = bm == BM && !bn128 && (CRS >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(K, BN) * ksplit < 768);
   int nk = cdiv(CRS, use64 ? 64 : 32);
-  dim3 grid(cdiv(M, bm), cdiv(K, BN), ksplit);
-  if (bm == 256)
+  dim3 grid(cdiv(M, bm), cdiv(K, bn128 ? 128 : BN), ksplit);
+  if (bn128)
+    hipLaunchKernelGGL((conv2d_fwd_kernel<32, 128, 128>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,
+                       (float*)ws, g, M, CRS, nk, fast_a, ksplit);
+  else if (bm == 256)
     hipLaunchKernelGGL((conv2d_fwd_kernel<32, 256>), grid, dim3(NTHREADS), 0,
                        (hipStream_t)stream, (const bf16_t*)x,
                        (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,

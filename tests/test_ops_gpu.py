@@ -47,6 +47,7 @@ def test_extension_loaded():
     (64, 128, 1, 2, 8, 4),    # 1x1 stride-2 projection
     (3, 64, 7, 2, 32, 2),     # imagenet stem
     (48, 32, 3, 1, 8, 2),     # C not mult of 32 -> generic path
+    (128, 128, 3, 1, 16, 128),  # big-M K=128 -> BN=128 fwd tile path
 ])
 def test_conv_fwd_bwd(cin, cout, k, stride, hw, batch):
     pad = k // 2
