@@ -322,3 +322,35 @@ def test_wa_loss_gpu():
     outg[0].backward()
     outc[0].backward()
     _cmp(sg.grad, sc.grad, rtol=2e-2, atol=2e-3, what="wa_loss grad")
+
+
+def test_conv_shape_fuzz():
+    """Irregular geometries (odd spatial, ragged M/K tails, stride+pad combos)
+    vs the CPU oracle — guards the staging bounds/padding logic."""
+    torch.manual_seed(99)
+    shapes = [
+        # (B, H, W, Cin, Cout, k, stride, pad)
+        (3, 7, 7, 16, 16, 3, 1, 1),
+        (5, 9, 9, 32, 48, 3, 2, 1),
+        (2, 11, 11, 64, 24, 1, 1, 0),
+        (7, 5, 5, 24, 64, 3, 1, 1),
+        (1, 17, 17, 16, 40, 5, 2, 2),
+        (4, 6, 6, 128, 72, 3, 1, 1),
+        (2, 14, 14, 40, 16, 7, 2, 3),
+    ]
+    for (B, H, W, Ci, Co, k, st, pd) in shapes:
+        x = torch.randn(B, H, W, Ci).to(torch.bfloat16)
+        w = (torch.randn(k, k, Ci, Co) * 0.2)
+        xg = x.cuda().requires_grad_()
+        xc = x.clone().requires_grad_()
+        wg = w.clone().cuda().requires_grad_()
+        wc = w.clone().requires_grad_()
+        yg = CF.conv2d(xg, wg, st, pd)
+        yc = CF.conv2d(xc, wc, st, pd)
+        _cmp(yg, yc, rtol=0.03, atol=0.03,
+             what=f"fuzz fwd {B}x{H}x{W}x{Ci}->{Co} k{k}s{st}p{pd}")
+        dy = torch.randn(yc.shape).to(torch.bfloat16)
+        yg.backward(dy.cuda())
+        yc.backward(dy)
+        _cmp(xg.grad, xc.grad, rtol=0.05, atol=0.05, what="fuzz dx")
+        _cmp(wg.grad, wc.grad, rtol=0.05, atol=0.05, what="fuzz dw")
