@@ -8,7 +8,6 @@ without torch internals.
 
 import math
 
-import numpy as np
 import torch
 
 

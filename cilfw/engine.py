@@ -19,7 +19,6 @@ from torch.utils.data import DataLoader
 from . import ops
 from .data import build_dataset, DistributedSampler, DATASET_STATS
 from .data.gpu_pipeline import GpuTaskLoader
-from .data.transforms import EvalTransform
 from .distributed import (init_distributed_mode, DataParallelEngine, barrier,
                           get_world_size, get_rank)
 from .models import CilModel

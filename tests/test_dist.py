@@ -3,7 +3,6 @@ metric sync — the CPU stand-in for the RCCL/xGMI path."""
 
 import os
 
-import numpy as np
 import pytest
 import torch
 import torch.distributed as dist

@@ -3,10 +3,7 @@
 Uses the class-separable synthetic dataset so a short run must actually learn;
 checks the acc1s trajectory shape, WA invocation, determinism and resume."""
 
-import copy
-
 import pytest
-import torch
 
 from cilfw.config import parse_args
 from cilfw.engine import run

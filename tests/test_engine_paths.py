@@ -6,7 +6,7 @@ import sys
 
 import pytest
 
-from cilfw.config import parse_args, get_args_parser
+from cilfw.config import parse_args
 from cilfw.engine import run
 
 

@@ -1,5 +1,4 @@
 import torch
-import pytest
 
 from cilfw.models import CilModel, get_backbone
 

@@ -4,7 +4,6 @@ Every kernel is compared against the same op computed by the CPU reference
 implementation on the SAME bf16-quantized inputs (asymmetric random data — a
 transposed MFMA C-write or swapped operand cannot pass, guide §5.4 rule 16)."""
 
-import numpy as np
 import pytest
 import torch
 

@@ -1,5 +1,4 @@
 import numpy as np
-import pytest
 
 from cilfw.data import (ClassIncremental, make_synthetic, CIFAR100_CLASS_ORDER,
                         build_dataset)
