@@ -94,6 +94,10 @@ def get_args_parser():
                         help="HBM-resident task data + on-device batch "
                              "assembly (crop/flip/normalize) — bypasses the "
                              "Python DataLoader for array-backed datasets")
+    parser.add_argument("--metric_every", default=1, type=int,
+                        help="read train metrics to host every N steps "
+                             "(1 = reference-exact; higher avoids per-step "
+                             "GPU syncs in the hot loop)")
     parser.add_argument("--no_step_graph", action="store_true", default=False,
                         help="disable hipGraph capture of the training step "
                              "(graphs are on by default with --gpu_data)")

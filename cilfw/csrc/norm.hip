@@ -257,8 +257,39 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
   __syncthreads();
   const float rM = 1.f / (float)M;
   long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
-  long iend = min(i0 + 8, total);
-  for (long i = i0; i < iend; ++i) {
+  if (i0 >= total) return;
+  if (i0 + 8 <= total) {  // vectorized b128 path (C % 8 == 0)
+    int4 dv = *(const int4*)&dy[i0];
+    int4 xv = *(const int4*)&x[i0];
+    int4 yv;
+    if (relu) yv = *(const int4*)&y[i0];
+    const bf16_t* de = (const bf16_t*)&dv;
+    const bf16_t* xe = (const bf16_t*)&xv;
+    const bf16_t* ye = (const bf16_t*)&yv;
+    __align__(16) bf16_t odx[8];
+    __align__(16) bf16_t ores[8];
+    int c0 = (int)(i0 % C);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float g = bf2f(de[j]);
+      if (relu && bf2f(ye[j]) <= 0.f) g = 0.f;
+      ores[j] = f2bf(g);
+      float v;
+      if (training) {
+        float xhat = (bf2f(xe[j]) - params[C + c]) * params[2 * C + c];
+        v = params[c] * rM *
+            ((float)M * g - params[4 * C + c] - xhat * params[3 * C + c]);
+      } else {
+        v = g * params[c];
+      }
+      odx[j] = f2bf(v);
+    }
+    *(int4*)&dx[i0] = *(int4*)odx;
+    if (dres != nullptr) *(int4*)&dres[i0] = *(int4*)ores;
+    return;
+  }
+  for (long i = i0; i < total; ++i) {
     int c = (int)(i % C);
     float g = bf2f(dy[i]);
     if (relu && bf2f(y[i]) <= 0.f) g = 0.f;

@@ -166,6 +166,8 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         nimg = 0
         graphed = None  # per-epoch: the cosine LR is baked into the capture
         first_of_epoch = True
+        metric_every = max(int(getattr(args, "metric_every", 1)), 1)
+        step_i = 0
         for inputs, targets, _tids in train_loader:
             inputs, targets = _to_device(inputs, targets, device, dtype)
             if can_graph and graphed is None and not first_of_epoch:
@@ -189,14 +191,16 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
                 if args.compat_step_barrier:
                     barrier()  # reference per-step barrier (template.py:272)
             first_of_epoch = False
-            accs = ops.accuracy(logits, targets,
-                                topk=(1, min(5, logits.shape[1])))
             bs = targets.shape[0]
             nimg += bs
-            metric_logger.update(ce=loss_ce.item(), kd=loss_kd.item(),
-                                 loss=loss.item())
-            metric_logger.update_n(n=bs, acc1=accs[0])
-            metric_logger.meters["lr"].update(optimizer.lr)
+            if step_i % metric_every == 0:  # host sync point (reads scalars)
+                accs = ops.accuracy(logits, targets,
+                                    topk=(1, min(5, logits.shape[1])))
+                metric_logger.update(ce=loss_ce.item(), kd=loss_kd.item(),
+                                     loss=loss.item())
+                metric_logger.update_n(n=bs, acc1=accs[0])
+                metric_logger.meters["lr"].update(optimizer.lr)
+            step_i += 1
         epoch_range.__exit__(None, None, None)
         metric_logger.synchronize_between_processes(device=torch.device(device))
         scheduler.step()

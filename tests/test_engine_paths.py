@@ -53,3 +53,17 @@ def test_template_entrypoint_help():
                          capture_output=True, text=True, timeout=120)
     assert out.returncode == 0
     assert "--num_bases" in out.stdout
+
+
+@pytest.mark.timeout(900)
+def test_imagenet_config_paths_and_metric_every():
+    """BASELINE configs 3/5 plumbing (synthetic stand-ins) + sampled metrics."""
+    args = parse_args([
+        "--data_set", "imagenet100", "--backbone", "resnet18",
+        "--num_bases", "50", "--increment", "50", "--num_epochs", "1",
+        "--batch_size", "16", "--workers", "0", "--input_size", "32",
+        "--no_aug", "--memory_size", "40", "--eval_every_epoch", "0",
+        "--metric_every", "4",
+    ])
+    accs = run(args)
+    assert len(accs) == 2
