@@ -38,6 +38,7 @@ _PROTOS = {
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 12 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
+    "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
     "cilfw_bn_fwd": [c_vp] * 10 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
     "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
@@ -209,6 +210,21 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
     _bf16(x, "bn_fwd.x")
     if residual is not None:
         _bf16(residual, "bn_fwd.residual")
+    frozen = (None if training
+              else getattr(running_mean, "_cilfw_frozen", None))
+    if frozen is not None:
+        mean, invstd = frozen
+        C = x.shape[-1]
+        M = x.numel() // C
+        y = torch.empty_like(x)
+        gf = gamma.float().contiguous()
+        bf = beta.float().contiguous()
+        _lib.cilfw_bn_apply_only(_ptr(x), _ptr(y), _ptr(residual), _ptr(gf),
+                                 _ptr(bf), _ptr(mean), _ptr(invstd),
+                                 c_l(M * C), c_i(C),
+                                 c_i(1 if relu else 0), _stream())
+        _check("bn_apply_only")
+        return y, mean, invstd
     C = x.shape[-1]
     assert C % 8 == 0, "bn kernels vectorize over channels (C%8==0)"
     M = x.numel() // C

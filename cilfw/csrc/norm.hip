@@ -83,6 +83,40 @@ void bn_reduce_slabs_kernel(const float* __restrict__ ws,
   if (lane == 0) out[i] = a;
 }
 
+// fused slab-reduce + finalize for TRAINING: one wave per channel reduces the
+// [gy][2][C] partials and computes mean/invstd/running-stat update directly
+// (drops one launch per training BN forward).
+__global__ __launch_bounds__(NT)
+void bn_reduce_finalize_kernel(const float* __restrict__ part,
+                               float* __restrict__ mean,
+                               float* __restrict__ invstd,
+                               float* __restrict__ running_mean,
+                               float* __restrict__ running_var,
+                               int gy, long M, int C, float momentum,
+                               float eps) {
+  const int lane = threadIdx.x & 63;
+  const int c = blockIdx.x * (NT / WAVE) + (threadIdx.x >> 6);
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int g = lane; g < gy; g += WAVE) {
+    s += part[((long)g * 2) * C + c];
+    q += part[((long)g * 2 + 1) * C + c];
+  }
+  for (int o = 32; o > 0; o >>= 1) {
+    s += __shfl_xor(s, o);
+    q += __shfl_xor(q, o);
+  }
+  if (lane == 0) {
+    float mu = s / (float)M;
+    float var = fmaxf(q / (float)M - mu * mu, 0.f);
+    mean[c] = mu;
+    invstd[c] = rsqrtf(var + eps);
+    float unbiased = var * (float)M / (float)max(M - 1, 1L);
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
 // pass 2: finalize mean/invstd (+ running stats update, training only)
 __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    const float* __restrict__ sumsq,
@@ -545,6 +579,20 @@ void cilfw_bn_fwd(const void* x, void* y, const void* res,
                      (const bf16_t*)res, (const float*)gamma,
                      (const float*)beta, (const float*)mean,
                      (const float*)invstd, total, C, relu);
+}
+
+void cilfw_bn_apply_only(const void* x, void* y, const void* res,
+                         const void* gamma, const void* beta,
+                         const void* mean, const void* invstd, long total,
+                         int C, int relu, void* stream) {
+  // frozen-model eval: stats precomputed, apply is the only launch
+  long blocks = cdiv((long)total, (long)NT * 8);
+  hipLaunchKernelGGL(bn_apply_kernel, dim3((int)blocks), dim3(NT),
+                     4 * C * sizeof(float), (hipStream_t)stream,
+                     (const bf16_t*)x, (bf16_t*)y, (const bf16_t*)res,
+                     (const float*)gamma, (const float*)beta,
+                     (const float*)mean, (const float*)invstd, total, C,
+                     relu);
 }
 
 void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,

@@ -119,11 +119,18 @@ class CilModel(nn.Module):
     def cast_compute_weights_(self, dtype):
         """Convert conv/linear weights to the compute dtype in place — used on
         the FROZEN teacher so its forward skips the per-step fp32->bf16 casts
-        (the fp32 masters only matter for models that train)."""
-        from .layers import Conv2d
+        (the fp32 masters only matter for models that train). Also precomputes
+        each BN's eval (mean, invstd) so the per-call finalize launch is
+        skipped (the running stats of a frozen model never change)."""
+        from .layers import Conv2d, BatchNormAct2d
         for m in self.modules():
             if isinstance(m, Conv2d):
                 m.weight.data = m.weight.data.to(dtype)
+            elif isinstance(m, BatchNormAct2d):
+                mean = m.running_mean.float().clone()
+                invstd = torch.rsqrt(m.running_var.float() + m.eps)
+                m.running_mean._cilfw_frozen = (mean.contiguous(),
+                                                invstd.contiguous())
         if self.fc is not None:
             for h in self.fc.heads:
                 h.weight.data = h.weight.data.to(dtype)
