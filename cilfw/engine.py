@@ -149,9 +149,12 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
             optimizer.step()
         return logits, loss_ce, loss_kd, loss
 
+    import os as _os
     can_graph = (str(device).startswith("cuda") and not args.no_step_graph
                  and not args.compat_step_barrier
-                 and not isinstance(train_loader, DataLoader))
+                 and not isinstance(train_loader, DataLoader)
+                 and (get_world_size() == 1
+                      or _os.environ.get("CILFW_GRAPH_MULTI") == "1"))
     model.train()
     for epoch in range(args.num_epochs):
         if train_sampler is not None:
