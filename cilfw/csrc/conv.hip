@@ -461,6 +461,168 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       }
 }
 
+// ==================== backward data, stride-2 fused parity ====================
+// dX positions partition by (hi%2, wi%2) into 4 classes, each touched by only
+// the (r,s) with r = r0+2*jr, s = s0+2*js — one launch, blockIdx.z = class,
+// removes the 4x zero-structured MFMA work of the generic strided gather
+// without extra launches. Requires stride == 2 and K % 16 == 0 (fast gather).
+
+template <int BKT>
+__global__ __launch_bounds__(NTHREADS)
+void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
+                               const bf16_t* __restrict__ w,
+                               bf16_t* __restrict__ dx, ConvGeom g) {
+  constexpr int NQ = BKT / 32;
+  constexpr int LPX = BKT + 8;
+  __shared__ bf16_t lds[2 * BM * LPX + 2 * BN * LPX];
+  const int AS0 = 0, BS0 = 2 * BM * LPX;
+  const int cls = blockIdx.z;
+  const int ph = cls >> 1, pw = cls & 1;
+  const int H2 = (g.H - ph + 1) >> 1, W2 = (g.W - pw + 1) >> 1;
+  const int r0 = (ph + g.pad) & 1, s0 = (pw + g.pad2) & 1;
+  const int nr = (g.R - r0 + 1) >> 1, ns = (g.S - s0 + 1) >> 1;
+  const int padh = (ph + g.pad - r0) >> 1, padw = (pw + g.pad2 - s0) >> 1;
+  const int Mc = g.N * H2 * W2;
+  const int RSK = nr * ns * g.K;
+  const int nk = cdiv_i(RSK, BKT);
+  const int m0 = blockIdx.x * BM;
+  if (m0 >= Mc || nr <= 0 || ns <= 0) return;
+  const int c0 = blockIdx.y * BN;
+  const int t = threadIdx.x;
+  const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
+
+  const int arow = t >> 1, ahalf = t & 1;
+  int m = m0 + arow;
+  int an = 0, ah2 = 0, aw2 = 0;
+  bool arow_ok = m < Mc;
+  if (arow_ok) {
+    an = m / (H2 * W2);
+    int rem = m - an * (H2 * W2);
+    ah2 = rem / W2;
+    aw2 = rem - ah2 * W2;
+  }
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
+
+  int4 areg[2 * NQ];
+  __align__(16) bf16_t breg[NQ][8];
+
+  // incremental (jr, js, kc) decomposition per q-chunk
+  int inc_jr[NQ], inc_js[NQ], inc_kc[NQ];
+#pragma unroll
+  for (int q = 0; q < NQ; ++q) {
+    int k = q * 32 + ahalf * 16;  // kt0 == 0 (no ksplit)
+    int rs = k / g.K;
+    inc_kc[q] = k - rs * g.K;
+    inc_jr[q] = rs / ns;
+    inc_js[q] = rs - inc_jr[q] * ns;
+  }
+
+  auto stage_to_regs = [&](int kt) {
+    const int k0 = kt * BKT;
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      int kc0 = inc_kc[q], jr = inc_jr[q], js = inc_js[q];
+      {
+        int ho = ah2 + padh - jr, wo = aw2 + padw - js;
+        if (arow_ok && ho >= 0 && ho < g.Ho && wo >= 0 && wo < g.Wo) {
+          const int4* src = (const int4*)&dy[(((long)an * g.Ho + ho) * g.Wo
+                                              + wo) * g.K + kc0];
+          areg[2 * q] = src[0];
+          areg[2 * q + 1] = src[1];
+        } else {
+          areg[2 * q] = int4{0, 0, 0, 0};
+          areg[2 * q + 1] = int4{0, 0, 0, 0};
+        }
+        kc0 += BKT;
+        while (kc0 >= g.K) {
+          kc0 -= g.K;
+          if (++js == ns) { js = 0; ++jr; }
+        }
+        inc_kc[q] = kc0; inc_jr[q] = jr; inc_js[q] = js;
+      }
+      // B: Bs[c][kk] = w[((r0+2jr)*S + s0+2js)*C + c][kc]
+      const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
+      int k = k0 + bkk;
+      if (k < RSK && c0 + bc < g.C) {
+        int rs = k / g.K, kc = k - rs * g.K;
+        int jr2 = rs / ns, js2 = rs - jr2 * ns;
+        int rs_orig = (r0 + 2 * jr2) * g.S + s0 + 2 * js2;
+        *(int4*)breg[q] = *(const int4*)&w[((long)rs_orig * g.C + c0 + bc)
+                                           * g.K + kc];
+      } else {
+        *(int4*)breg[q] = int4{0, 0, 0, 0};
+      }
+    }
+  };
+
+  auto regs_to_lds = [&](int buf) {
+    bf16_t* As = &lds[AS0 + buf * BM * LPX];
+    bf16_t* Bs = &lds[BS0 + buf * BN * LPX];
+#pragma unroll
+    for (int q = 0; q < NQ; ++q) {
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16] = areg[2 * q];
+      *(int4*)&As[arow * LPX + q * 32 + ahalf * 16 + 8] = areg[2 * q + 1];
+      const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
+      *(int4*)&Bs[bc * LPX + bkk] = *(int4*)breg[q];
+    }
+  };
+
+  stage_to_regs(0);
+  regs_to_lds(0);
+  __syncthreads();
+  FragIdx fi = frag_idx();
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    if (kt + 1 < nk) stage_to_regs(kt + 1);
+    {
+      const bf16_t* As = &lds[AS0 + cur * BM * LPX];
+      const bf16_t* Bs = &lds[BS0 + cur * BN * LPX];
+#pragma unroll
+      for (int q = 0; q < NQ; ++q) {
+        const int kb = fi.quad * 8 + q * 32;
+#pragma unroll
+        for (int mr = 0; mr < 4; ++mr) {
+          int row = wr * 64 + mr * 16 + fi.half;
+          bf16x8 a = *(const bf16x8*)&As[row * LPX + kb];
+#pragma unroll
+          for (int nrr = 0; nrr < 2; ++nrr) {
+            int col = wc * 32 + nrr * 16 + fi.half;
+            bf16x8 b = *(const bf16x8*)&Bs[col * LPX + kb];
+            acc[mr][nrr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a, b, acc[mr][nrr], 0, 0, 0);
+          }
+        }
+      }
+    }
+    if (kt + 1 < nk) {
+      regs_to_lds(cur ^ 1);
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+    for (int nrr = 0; nrr < 2; ++nrr)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mr * 16 + fi.quad * 4 + r;
+        int col = c0 + wc * 32 + nrr * 16 + fi.half;
+        if (row < Mc && col < g.C) {
+          int n = row / (H2 * W2);
+          int rem = row - n * (H2 * W2);
+          int h2 = rem / W2, w2 = rem - h2 * W2;
+          dx[(((long)n * g.H + ph + 2 * h2) * g.W + pw + 2 * w2) * g.C + col]
+              = f2bf(acc[mr][nrr][r]);
+        }
+      }
+}
+
 // ============================== backward weight ==============================
 // dW[(r,s,c), k] = sum_m X[m -> (n,hi,wi,c)] * dY[m, k]; the m-reduction is
 // split over blockIdx.z into fp32 partial slabs (reduced by reduce_slabs_f32).
@@ -745,6 +907,17 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
                            int stride, int pad, int Ho, int Wo, int ksplit,
                            void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
+  if (stride == 2 && (R > 1 || S > 1) && K % 16 == 0 &&
+      getenv("CILFW_NO_S2_FUSED") == nullptr) {
+    // fused 4-class parity decomposition: grid.z = (hi%2, wi%2) class
+    int H2max = (H + 1) >> 1, W2max = (W + 1) >> 1;
+    int Mc = N * H2max * W2max;
+    dim3 grid(cdiv(Mc, BM), cdiv(C, BN), 4);
+    hipLaunchKernelGGL((conv2d_bwd_data_s2_kernel<32>), grid, dim3(NTHREADS),
+                       0, (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, g);
+    return;
+  }
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);
