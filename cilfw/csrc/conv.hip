@@ -907,11 +907,14 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
                            int stride, int pad, int Ho, int Wo, int ksplit,
                            void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
+  int H2max = (H + 1) >> 1, W2max = (W + 1) >> 1;
+  int Mc = N * H2max * W2max;
   if (stride == 2 && (R > 1 || S > 1) && K % 16 == 0 &&
+      cdiv(Mc, BM) * cdiv(C, BN) >= 128 &&
       getenv("CILFW_NO_S2_FUSED") == nullptr) {
-    // fused 4-class parity decomposition: grid.z = (hi%2, wi%2) class
-    int H2max = (H + 1) >> 1, W2max = (W + 1) >> 1;
-    int Mc = N * H2max * W2max;
+    // fused 4-class parity decomposition: grid.z = (hi%2, wi%2) class;
+    // small per-class grids fall back to the single zero-structured kernel
+    // (its ksplit fills the chip better there — measured)
     dim3 grid(cdiv(Mc, BM), cdiv(C, BN), 4);
     hipLaunchKernelGGL((conv2d_bwd_data_s2_kernel<32>), grid, dim3(NTHREADS),
                        0, (hipStream_t)stream, (const bf16_t*)dy,
