@@ -51,20 +51,50 @@ def _to_device(inputs, targets, device, dtype):
 
 @torch.no_grad()
 def evaluate(model, loader, device, args, header="Test:"):
+    """Cumulative eval (reference template.py:169-188). Correct counts and the
+    loss sum accumulate ON DEVICE (exact, not sampled) — one host sync at the
+    end instead of per batch."""
     model.eval()
     dtype = compute_dtype(args)
-    metric_logger = MetricLogger()
     nb_classes = model.fc.nb_classes
-    topk = (1, min(5, nb_classes))
+    kmax = min(5, nb_classes)
+    on_dev = str(device).startswith("cuda")
+    dev = torch.device(device)
+    correct = torch.zeros(2, dtype=torch.float64, device=dev)
+    loss_sum = torch.zeros((), dtype=torch.float64, device=dev)
+    n_total = 0
+    n_batches = 0
     for inputs, targets, _tids in loader:
         inputs, targets = _to_device(inputs, targets, device, dtype)
         logits, _ = model(inputs)
         loss, _ce, _kd = ops.wa_loss(logits, None, targets)
-        accs = ops.accuracy(logits, targets, topk=topk)
         bs = targets.shape[0]
-        metric_logger.update(loss=loss.item())
-        metric_logger.update_n(n=bs, acc1=accs[0], acc5=accs[1])
-    metric_logger.synchronize_between_processes(device=torch.device(device))
+        if on_dev:
+            from .ops._backend import use_hip, ext
+            if use_hip(logits):
+                counts = ext().topk_correct(logits, targets, kmax)
+                correct[0] += counts[0]
+                correct[1] += counts[kmax - 1]
+            else:
+                a = ops.accuracy(logits, targets, topk=(1, kmax))
+                correct[0] += a[0] * bs / 100.0
+                correct[1] += a[1] * bs / 100.0
+        else:
+            a = ops.accuracy(logits, targets, topk=(1, kmax))
+            correct[0] += a[0] * bs / 100.0
+            correct[1] += a[1] * bs / 100.0
+        loss_sum += loss.double()
+        n_total += bs
+        n_batches += 1
+    # exact cross-rank sample-weighted means (reference utils.py:36-43)
+    metric_logger = MetricLogger()
+    metric_logger.update_n(n=n_total, acc1=correct[0].item() * 100.0
+                           / max(n_total, 1))
+    metric_logger.update_n(n=n_total, acc5=correct[1].item() * 100.0
+                           / max(n_total, 1))
+    metric_logger.update_n(n=n_batches, loss=loss_sum.item()
+                           / max(n_batches, 1))
+    metric_logger.synchronize_between_processes(device=dev)
     acc1 = metric_logger.meters["acc1"].global_avg
     acc5 = metric_logger.meters["acc5"].global_avg
     lossv = metric_logger.meters["loss"].global_avg
