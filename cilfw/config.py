@@ -94,6 +94,9 @@ def get_args_parser():
                         help="HBM-resident task data + on-device batch "
                              "assembly (crop/flip/normalize) — bypasses the "
                              "Python DataLoader for array-backed datasets")
+    parser.add_argument("--max_tasks", default=0, type=int,
+                        help="stop after N tasks (0 = all) — e.g. the "
+                             "BASELINE config[0] 2-task plumbing oracle")
     parser.add_argument("--metric_every", default=1, type=int,
                         help="read train metrics to host every N steps "
                              "(1 = reference-exact; higher avoids per-step "

@@ -302,7 +302,10 @@ def run(args):
         print(f"resumed from {args.resume}: start_task={start_task}, "
               f"known={args.known_classes}, acc1s={acc1s}")
 
-    for task_id in range(start_task, len(scenario_train)):
+    nb_tasks = len(scenario_train)
+    if getattr(args, "max_tasks", 0):
+        nb_tasks = min(nb_tasks, args.max_tasks)
+    for task_id in range(start_task, nb_tasks):
         args.task_id = task_id
         dataset_train = scenario_train[task_id]
         dataset_val = scenario_val[:task_id + 1]

@@ -67,3 +67,19 @@ def test_imagenet_config_paths_and_metric_every():
     ])
     accs = run(args)
     assert len(accs) == 2
+
+
+@pytest.mark.timeout(900)
+def test_baseline_config0_b0inc10_resnet32():
+    """BASELINE.json config[0]: CIFAR-100-shaped B0-Inc10 ResNet-32 on CPU,
+    world_size=1, first 2 tasks only."""
+    args = parse_args([
+        "--data_set", "synthetic", "--backbone", "resnet32",
+        "--num_bases", "0", "--increment", "10", "--max_tasks", "2",
+        "--num_epochs", "1", "--batch_size", "32", "--workers", "0",
+        "--synthetic_train_size", "2000", "--memory_size", "200",
+        "--eval_every_epoch", "0", "--input_size", "32", "--no_aug",
+    ])
+    accs = run(args)
+    assert len(accs) == 2
+    assert args.known_classes == 20
