@@ -65,6 +65,7 @@ _PROTOS = {
     "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_f, c_f, c_f, c_vp],
     "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_herding_select": [c_vp] * 3 + [c_i] * 3 + [c_vp],
+    "cilfw_herding_select_batch": [c_vp] * 6 + [c_i] * 3 + [c_vp],
 }
 for _name, _args in _PROTOS.items():
     _fn = getattr(_lib, _name)
@@ -489,3 +490,32 @@ def herding_select(f, mu, m):
                               c_i(m), _stream())
     _check("herding_select")
     return order
+
+
+def herding_select_batch(feats_list, m_list):
+    """All classes' greedy herding in ONE launch (block per class).
+    feats_list: list of (n_c, D) fp32 GPU tensors; returns list of int64
+    ranked-index tensors (local indices, length m_c)."""
+    device = feats_list[0].device
+    D = feats_list[0].shape[1]
+    ns = [int(f.shape[0]) for f in feats_list]
+    assert all((n + D) * 4 <= 60_000 for n in ns), "herding LDS limit"
+    fall = torch.cat([f.float().contiguous() for f in feats_list])
+    mus = torch.stack([f.float().mean(0) for f in feats_list]).contiguous()
+    foff = torch.tensor([0] + list(torch.tensor(ns).cumsum(0)),
+                        dtype=torch.int32, device=device)
+    ms = [min(m, n) for m, n in zip(m_list, ns)]
+    ooff = torch.tensor([0] + list(torch.tensor(ms).cumsum(0)),
+                        dtype=torch.int32, device=device)
+    mvec = torch.tensor(ms, dtype=torch.int32, device=device)
+    order = torch.empty(sum(ms), dtype=torch.int64, device=device)
+    _lib.cilfw_herding_select_batch(_ptr(fall), _ptr(foff), _ptr(mus),
+                                    _ptr(ooff), _ptr(mvec), _ptr(order),
+                                    c_i(len(ns)), c_i(D), c_i(max(ns)),
+                                    _stream())
+    _check("herding_select_batch")
+    outs, o = [], 0
+    for m in ms:
+        outs.append(order[o:o + m])
+        o += m
+    return outs

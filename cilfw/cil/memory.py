@@ -79,11 +79,26 @@ class RehearsalMemory:
         new_classes = [c for c in np.unique(y) if c not in self._x]
         nb_seen = self.nb_classes + len(new_classes)
         quota = self._quota(nb_seen)
+        idxs = {c: np.where(y == c)[0] for c in new_classes}
+        ranked_by_class = {}
+        if self.herding_method == "barycenter" and features.is_cuda \
+                and len(new_classes) > 1:
+            # all classes in ONE launch — concurrent per-class greedy loops
+            from ..ops._backend import use_hip, ext
+            if use_hip(features):
+                feats = [features[idxs[c]] for c in new_classes]
+                # rank only the kept prefix (greedy prefix property: the
+                # first q picks are identical whatever the total m)
+                orders = ext().herding_select_batch(
+                    feats, [quota for _ in new_classes])
+                for c, o in zip(new_classes, orders):
+                    ranked_by_class[c] = o.cpu().numpy()
         for c in new_classes:
-            idx = np.where(y == c)[0]
-            feats_c = features[idx]
-            if self.herding_method == "barycenter":
-                ranked = herding_select(feats_c, len(idx)).cpu().numpy()
+            idx = idxs[c]
+            if c in ranked_by_class:
+                ranked = ranked_by_class[c]
+            elif self.herding_method == "barycenter":
+                ranked = herding_select(features[idx], quota).cpu().numpy()
             elif self.herding_method == "random":
                 rng = np.random.default_rng(int(c))
                 ranked = rng.permutation(len(idx))

@@ -450,6 +450,75 @@ void herding_kernel(const float* __restrict__ f, const float* __restrict__ mu,
   }
 }
 
+// batched herding: one BLOCK per class, all classes' greedy selections run
+// CONCURRENTLY (the per-class loop is inherently sequential in m, so the
+// parallelism axis is classes — SURVEY §7 hard-parts item). Layout: features
+// flat (sum n_c, D) grouped by class with offsets foff[c]; per-class means
+// mu[c]; outputs ranked indices (LOCAL to the class) at ooff[c].
+__global__ __launch_bounds__(NT)
+void herding_batch_kernel(const float* __restrict__ fall,
+                          const int* __restrict__ foff,
+                          const float* __restrict__ muall,
+                          const int* __restrict__ ooff,
+                          const int* __restrict__ mvec,
+                          long* __restrict__ order, int D) {
+  const int cls = blockIdx.x;
+  const float* f = fall + (long)foff[cls] * D;
+  const float* mu = muall + (long)cls * D;
+  long* out = order + ooff[cls];
+  const int n = foff[cls + 1] - foff[cls];
+  const int m = mvec[cls];
+  extern __shared__ float sh[];
+  float* sum_sel = sh;
+  float* selmark = sh + D;
+  for (int d = threadIdx.x; d < D; d += NT) sum_sel[d] = 0.f;
+  for (int i = threadIdx.x; i < n; i += NT) selmark[i] = 0.f;
+  __shared__ float rv[NT / WAVE];
+  __shared__ int ri[NT / WAVE];
+  __shared__ int chosen;
+  __syncthreads();
+  for (int k = 0; k < m; ++k) {
+    const float inv = 1.f / (k + 1);
+    float best = 3.4e38f;
+    int bi = INT32_MAX;
+    for (int i = threadIdx.x; i < n; i += NT) {
+      if (selmark[i] != 0.f) continue;
+      const float* fi = f + (long)i * D;
+      float d2 = 0.f;
+      for (int d = 0; d < D; ++d) {
+        float t = (sum_sel[d] + fi[d]) * inv - mu[d];
+        d2 += t * t;
+      }
+      if (d2 < best || (d2 == best && i < bi)) { best = d2; bi = i; }
+    }
+    for (int o = 32; o > 0; o >>= 1) {
+      float ov = __shfl_xor(best, o);
+      int oi = __shfl_xor(bi, o);
+      if (ov < best || (ov == best && oi < bi)) { best = ov; bi = oi; }
+    }
+    __syncthreads();
+    if ((threadIdx.x & 63) == 0) {
+      rv[threadIdx.x >> 6] = best;
+      ri[threadIdx.x >> 6] = bi;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int w = 1; w < NT / WAVE; ++w)
+        if (rv[w] < best || (rv[w] == best && ri[w] < bi)) {
+          best = rv[w];
+          bi = ri[w];
+        }
+      chosen = bi;
+      out[k] = bi;
+      selmark[bi] = 1.f;
+    }
+    __syncthreads();
+    const float* fc = f + (long)chosen * D;
+    for (int d = threadIdx.x; d < D; d += NT) sum_sel[d] += fc[d];
+    __syncthreads();
+  }
+}
+
 // ============================== launchers ==============================
 
 extern "C" {
@@ -509,6 +578,17 @@ void cilfw_topk_correct(const void* logits, const void* targets, void* counts,
   hipLaunchKernelGGL(topk_kernel, dim3(M), dim3(NT), 0, (hipStream_t)stream,
                      (const float*)logits, (const long*)targets,
                      (long*)counts, M, C, maxk);
+}
+
+void cilfw_herding_select_batch(const void* fall, const void* foff,
+                                const void* mu, const void* ooff,
+                                const void* mvec, void* order, int nclasses,
+                                int D, int max_n, void* stream) {
+  size_t shmem = (size_t)(D + max_n) * sizeof(float);
+  hipLaunchKernelGGL(herding_batch_kernel, dim3(nclasses), dim3(NT), shmem,
+                     (hipStream_t)stream, (const float*)fall,
+                     (const int*)foff, (const float*)mu, (const int*)ooff,
+                     (const int*)mvec, (long*)order, D);
 }
 
 void cilfw_herding_select(const void* f, const void* mu, void* order, int n,

@@ -353,3 +353,15 @@ def test_conv_shape_fuzz():
         yc.backward(dy)
         _cmp(xg.grad, xc.grad, rtol=0.05, atol=0.05, what="fuzz dx")
         _cmp(wg.grad, wc.grad, rtol=0.05, atol=0.05, what="fuzz dw")
+
+
+def test_herding_batch_matches_single():
+    from cilfw.ops._backend import ext
+    torch.manual_seed(41)
+    feats = [torch.randn(120 + 13 * i, 32).cuda() for i in range(5)]
+    batch = ext().herding_select_batch(feats, [20] * 5)
+    for f, got in zip(feats, batch):
+        mu = f.mean(0)
+        single = ext().herding_select(f.float().contiguous(),
+                                      mu.contiguous(), 20)
+        assert torch.equal(got.cpu(), single.cpu())
