@@ -61,7 +61,11 @@ class CilClassifier(nn.Module):
         return sum(h.out_features for h in self.heads)
 
     def forward(self, x):
-        # one fused GEMM over concatenated head weights (heads are tiny: C x 64..2048)
+        # one fused GEMM over concatenated head weights (heads are tiny:
+        # C x 64..2048); frozen models use a cached concatenation
+        cat = getattr(self, "_frozen_cat", None)
+        if cat is not None:
+            return CF.linear(x, cat[0], cat[1])
         w = torch.cat([h.weight for h in self.heads], dim=0)
         b = torch.cat([h.bias for h in self.heads], dim=0)
         return CF.linear(x, w, b)
@@ -135,6 +139,9 @@ class CilModel(nn.Module):
             for h in self.fc.heads:
                 h.weight.data = h.weight.data.to(dtype)
                 h.bias.data = h.bias.data.to(dtype)
+            self.fc._frozen_cat = (
+                torch.cat([h.weight for h in self.fc.heads], 0).contiguous(),
+                torch.cat([h.bias for h in self.fc.heads], 0).contiguous())
         return self
 
     def prev_model_adaption(self, nb_classes):
