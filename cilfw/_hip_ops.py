@@ -38,6 +38,7 @@ _PROTOS = {
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 12 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
+    "cilfw_im2col_smallc": [c_vp] * 3 + [c_i] * 11 + [c_vp],
     "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
     "cilfw_bn_fwd": [c_vp] * 10 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
     "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
@@ -107,6 +108,22 @@ def _bf16(t, name):
 
 # ------------------------------------------------------------------------ conv
 
+def _stem_cols(x, stride, pad, R, S, CRSpad):
+    """Padded im2col for tiny-C stems -> (M, CRSpad) bf16 (pre-zeroed)."""
+    N, H, W_, C = x.shape
+    Ho = (H + 2 * pad - R) // stride + 1
+    Wo = (W_ + 2 * pad - S) // stride + 1
+    M = N * Ho * Wo
+    mt = _mtable(N, Ho, Wo, stride, x.device)
+    col = torch.zeros(M, CRSpad, dtype=torch.bfloat16, device=x.device)
+    _lib.cilfw_im2col_smallc(_ptr(x), _ptr(mt), _ptr(col), c_i(N), c_i(H),
+                             c_i(W_), c_i(C), c_i(R), c_i(S), c_i(stride),
+                             c_i(pad), c_i(Ho), c_i(Wo), c_i(CRSpad),
+                             _stream())
+    _check("im2col_smallc")
+    return col, Ho, Wo
+
+
 def conv2d_fwd(x, w, stride, pad):
     _bf16(x, "conv2d_fwd.x")
     _bf16(w, "conv2d_fwd.w")
@@ -116,6 +133,25 @@ def conv2d_fwd(x, w, stride, pad):
     assert K % 8 == 0, "conv kernels vector-stage over output channels (K%8==0)"
     Ho = (H + 2 * pad - R) // stride + 1
     Wo = (W_ + 2 * pad - S) // stride + 1
+    if C < 16 and R * S > 1 and K % 16 == 0:
+        # stem path: padded im2col (one gather pass), then the fast-path
+        # flat GEMM (the generic per-element gather loop is far slower)
+        CRS = C * R * S
+        CRSpad = (CRS + 15) // 16 * 16
+        col, Ho, Wo = _stem_cols(x, stride, pad, R, S, CRSpad)
+        M = N * Ho * Wo
+        wpad = torch.zeros(CRSpad, K, dtype=torch.bfloat16, device=x.device)
+        wpad[:CRS] = w.reshape(CRS, K)
+        y = torch.empty(M, K, dtype=torch.bfloat16, device=x.device)
+        ks = _lib.cilfw_conv2d_fwd_ksplit(M, CRSpad, K, 1, 1, 1, 1)
+        ws = (torch.empty(ks * M * K, dtype=torch.float32, device=x.device)
+              if ks > 1 else None)
+        _lib.cilfw_conv2d_fwd(_ptr(col), _ptr(wpad), _ptr(y), _ptr(ws),
+                              c_i(M), c_i(1), c_i(1), c_i(CRSpad), c_i(K),
+                              c_i(1), c_i(1), c_i(1), c_i(0), c_i(1), c_i(1),
+                              c_i(ks), _stream())
+        _check("conv2d_fwd_stem")
+        return y.view(N, Ho, Wo, K)
     y = torch.empty(N, Ho, Wo, K, dtype=torch.bfloat16, device=x.device)
     ks = _lib.cilfw_conv2d_fwd_ksplit(N, C, K, R, S, Ho, Wo)
     ws = (torch.empty(ks * N * Ho * Wo * K, dtype=torch.float32,
@@ -183,6 +219,25 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
+    if C < 16 and R * S > 1 and K % 16 == 0:
+        # stem path: rebuild the padded im2col, dW = col^T @ dy via the flat
+        # bwd-weight kernel (n = m directly; no pixel table)
+        CRS = C * R * S
+        CRSpad = (CRS + 15) // 16 * 16
+        col, _, _ = _stem_cols(x, stride, pad, R, S, CRSpad)
+        M = N * Ho * Wo
+        dyf = dy.reshape(M, K)
+        dwp = torch.empty(CRSpad, K, dtype=torch.float32, device=x.device)
+        ns = _lib.cilfw_conv2d_bwd_weight_nslices(M, CRSpad, K, 1, 1, 1, 1)
+        ws = torch.empty(ns * CRSpad * K, dtype=torch.float32,
+                         device=x.device)
+        _lib.cilfw_conv2d_bwd_weight(_ptr(dyf), _ptr(col), _ptr(None),
+                                     _ptr(dwp), _ptr(ws), c_i(M), c_i(1),
+                                     c_i(1), c_i(CRSpad), c_i(K), c_i(1),
+                                     c_i(1), c_i(1), c_i(0), c_i(1), c_i(1),
+                                     c_i(ns), _stream())
+        _check("conv2d_bwd_weight_stem")
+        return dwp[:CRS].reshape(R, S, C, K)
     if R == 1 and S == 1 and stride > 1:
         xg = torch.empty(N, Ho, Wo, C, dtype=torch.bfloat16, device=x.device)
         _lib.cilfw_stride_gather(_ptr(x), _ptr(xg), c_i(N), c_i(H), c_i(W_),

@@ -669,14 +669,17 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   __align__(16) bf16_t breg[2][8];
   const int HoWo = g.Ho * g.Wo;
 
+  const bool flat = (g.H == 1 && g.W == 1 && g.Ho == 1 && g.Wo == 1);
   auto gather_one = [&](int m, bf16_t* adst, bf16_t* bdst) {
     int n = 0, ho = 0, wo = 0;
     const bool m_ok = m < me;
-    if (m_ok) {
+    if (m_ok && !flat) {
       n = m / HoWo;
       int rem = m - n * HoWo;
       ho = rem / g.Wo;
       wo = rem - ho * g.Wo;
+    } else if (m_ok) {
+      n = m;  // flat im2col geometry: one "pixel" per row
     }
     if (fast_a) {
       int hi = ho * g.stride - g.pad + r_;
@@ -803,6 +806,30 @@ void parity_scatter_kernel(const bf16_t* __restrict__ dxs,
   int h2 = (int)(rest % H2);
   int n = (int)(rest / H2);
   dx[(((long)n * H + ph + 2 * h2) * W + pw + 2 * w2) * C + c] = dxs[i];
+}
+
+// tiny-C (stem) im2col: col (M, CRSpad) bf16, one thread per (m, r*S+s)
+// copying the C input channels; OOB pixels stay zero (buffer pre-zeroed).
+// Pixel decomposition comes from the packed mt table.
+__global__ __launch_bounds__(NTHREADS)
+void im2col_smallc_kernel(const bf16_t* __restrict__ x,
+                          const int* __restrict__ mt,
+                          bf16_t* __restrict__ col, ConvGeom g, int M,
+                          int CRSpad, long total) {
+  long i = (long)blockIdx.x * NTHREADS + threadIdx.x;
+  if (i >= total) return;
+  const int RS = g.R * g.S;
+  int rs = (int)(i % RS);
+  int m = (int)(i / RS);
+  const int v = mt[m];
+  const int n = v >> 20;
+  const int r = rs / g.S, s = rs - (rs / g.S) * g.S;
+  const int hi = ((v >> 10) & 1023) - g.pad + r;
+  const int wi = (v & 1023) - g.pad2 + s;
+  if (hi < 0 || hi >= g.H || wi < 0 || wi >= g.W) return;
+  const bf16_t* src = &x[(((long)n * g.H + hi) * g.W + wi) * g.C];
+  bf16_t* dst = &col[(long)m * CRSpad + rs * g.C];
+  for (int c = 0; c < g.C; ++c) dst[c] = src[c];
 }
 
 // packed im2col pixel table: mt[m] = n<<20 | (ho*stride)<<10 | (wo*stride)
@@ -1018,6 +1045,20 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
                      dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
                      dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
                      (float*)dw, nslices, len);
+}
+
+void cilfw_im2col_smallc(const void* x, const void* mt, void* col,
+                         int N, int H, int W, int C, int R, int S,
+                         int stride, int pad, int Ho, int Wo, int CRSpad,
+                         void* stream) {
+  ConvGeom g{N, H, W, C, 0, R, S, stride, pad, pad, Ho, Wo};
+  int M = N * Ho * Wo;
+  long total = (long)M * R * S;
+  hipLaunchKernelGGL(im2col_smallc_kernel,
+                     dim3((int)cdiv((long)total, (long)NTHREADS)),
+                     dim3(NTHREADS), 0, (hipStream_t)stream,
+                     (const bf16_t*)x, (const int*)mt, (bf16_t*)col, g, M,
+                     CRSpad, total);
 }
 
 void cilfw_fill_mtable(void* mt, int M, int HoWo, int Wo, int stride,
