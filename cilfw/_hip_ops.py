@@ -133,9 +133,10 @@ def conv2d_fwd(x, w, stride, pad):
     assert K % 8 == 0, "conv kernels vector-stage over output channels (K%8==0)"
     Ho = (H + 2 * pad - R) // stride + 1
     Wo = (W_ + 2 * pad - S) // stride + 1
-    if C < 16 and R * S > 1 and K % 16 == 0:
-        # stem path: padded im2col (one gather pass), then the fast-path
-        # flat GEMM (the generic per-element gather loop is far slower)
+    if C < 16 and R * S > 1 and K % 16 == 0 \
+            and os.environ.get("CILFW_STEM_IM2COL") == "1":
+        # measured: the generic gather already fills the chip at stem sizes —
+        # this alternate path is kept for experiments (CILFW_STEM_IM2COL=1)
         CRS = C * R * S
         CRSpad = (CRS + 15) // 16 * 16
         col, Ho, Wo = _stem_cols(x, stride, pad, R, S, CRSpad)
@@ -219,9 +220,8 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
-    if C < 16 and R * S > 1 and K % 16 == 0:
-        # stem path: rebuild the padded im2col, dW = col^T @ dy via the flat
-        # bwd-weight kernel (n = m directly; no pixel table)
+    if C < 16 and R * S > 1 and K % 16 == 0 \
+            and os.environ.get("CILFW_STEM_IM2COL") == "1":
         CRS = C * R * S
         CRSpad = (CRS + 15) // 16 * 16
         col, _, _ = _stem_cols(x, stride, pad, R, S, CRSpad)
