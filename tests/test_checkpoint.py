@@ -48,3 +48,25 @@ def test_checkpoint_roundtrip(tmp_path):
     mx2, my2, _ = mem2.get()
     assert (mx1 == mx2).all() and (my1 == my2).all()
     assert state["acc1s"] == [50.0, 40.0]
+
+
+def test_checkpoint_restores_rng(tmp_path):
+    model = CilModel("resnet20", 32)
+    model.prev_model_adaption(3)
+    args = _Args()
+    torch.manual_seed(7)
+    _ = torch.randn(3)  # advance
+    mem = RehearsalMemory(memory_size=4)
+    mem.add(np.zeros((4, 2, 2, 3), dtype=np.uint8),
+            np.array([0, 0, 1, 1]), np.zeros(4, dtype=np.int64),
+            torch.randn(4, 2))
+    path = save_task_checkpoint(str(tmp_path), 0, model, mem, [1.0], args)
+    expected_next = torch.randn(2)  # what the RNG yields after saving
+
+    torch.manual_seed(999)  # scramble
+    model2 = CilModel("resnet20", 32)
+    mem2 = RehearsalMemory()
+    args2 = _Args()
+    load_task_checkpoint(path, model2, mem2, args2, restore_rng=True)
+    got = torch.randn(2)
+    assert torch.equal(got, expected_next)

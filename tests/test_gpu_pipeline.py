@@ -48,3 +48,19 @@ def test_augment_preserves_stats():
     imgs, _, _ = next(iter(ld))
     assert imgs.shape == (64, 16, 16, 3)
     assert torch.isfinite(imgs).all()
+
+
+def test_drop_last_and_epoch_reshuffle():
+    ts = _taskset(n=50)
+    ld = GpuTaskLoader(ts, 16, "cpu", (0.5,) * 3, (0.25,) * 3, shuffle=True,
+                       seed=3, augment=False, drop_last=True,
+                       dtype=torch.float32)
+    ld.set_epoch(0)
+    e0 = torch.cat([lab for _, lab, _ in ld])
+    ld.set_epoch(1)
+    e1 = torch.cat([lab for _, lab, _ in ld])
+    assert len(e0) == 48 and len(e1) == 48  # 3 batches, drop_last
+    assert not torch.equal(e0, e1)  # epoch changes the permutation
+    ld.set_epoch(0)
+    e0b = torch.cat([lab for _, lab, _ in ld])
+    assert torch.equal(e0, e0b)  # deterministic per (seed, epoch)
