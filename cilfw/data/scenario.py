@@ -37,7 +37,8 @@ class TaskSet(Dataset):
         if self.transform is not None:
             img = self.transform(img)
         else:
-            img = torch.from_numpy(np.ascontiguousarray(img)).float().div_(255.0)
+            img = torch.from_numpy(
+                np.ascontiguousarray(img).copy()).float().div_(255.0)
         return img, int(self.y[i]), int(self.t[i])
 
     def add_samples(self, x, y, t):
