@@ -59,6 +59,11 @@ class RehearsalMemory:
         self._x = {}   # class -> uint8 (k,H,W,C)
         self._y = {}
         self._t = {}
+        # last add()'s selection: class -> global indices (into the x passed
+        # to add, ranked) — lets a device-resident mirror gather the same
+        # exemplars from an already-uploaded task tensor with no host traffic
+        self.last_selection = {}
+        self.last_quota = None
 
     @property
     def nb_classes(self):
@@ -79,6 +84,8 @@ class RehearsalMemory:
         new_classes = [c for c in np.unique(y) if c not in self._x]
         nb_seen = self.nb_classes + len(new_classes)
         quota = self._quota(nb_seen)
+        self.last_selection = {}
+        self.last_quota = quota
         idxs = {c: np.where(y == c)[0] for c in new_classes}
         ranked_by_class = {}
         if self.herding_method == "barycenter" and features.is_cuda \
@@ -105,6 +112,7 @@ class RehearsalMemory:
             else:
                 raise ValueError(self.herding_method)
             keep = idx[ranked[:quota]]
+            self.last_selection[int(c)] = keep
             self._x[int(c)] = np.ascontiguousarray(x[keep])
             self._y[int(c)] = y[keep]
             self._t[int(c)] = t[keep]

@@ -84,3 +84,74 @@ class DeviceExemplarStore:
             imgs[flip] = imgs[flip].flip(2)
         imgs = (imgs - self.mean) / self.std
         return imgs.to(dtype), self.labels[idx]
+
+
+class DeviceReplayMirror:
+    """Device-resident mirror of a RehearsalMemory's exemplar *images* —
+    the engine's default replay source under ``--gpu_data``.
+
+    The reference re-materializes replay on the host every task
+    (memory.get() -> numpy concat -> DataLoader, template.py:230-231); here
+    exemplars selected from an already-uploaded task tensor are gathered ON
+    DEVICE (``update`` with ``task_images``) and stay resident in HBM across
+    tasks, so the replay fraction of every batch never touches the host.
+    Content is kept exactly equal to the host RehearsalMemory (same herding
+    keep-indices via ``memory.last_selection``, same quota trimming, same
+    sorted-class concat order as ``memory.get()``), which the engine asserts.
+    """
+
+    def __init__(self, device):
+        self.device = device
+        self._imgs = {}    # class -> uint8 (k,H,W,C) device tensor, ranked
+        self._labels = {}  # class -> int64 (k,) device tensor
+        self._tasks = {}   # class -> int64 (k,) device tensor
+
+    def update(self, memory, task_images=None, task_id=0):
+        """Ingest memory's last add(): gather new-class exemplars from the
+        device-resident ``task_images`` (uint8, aligned with the x passed to
+        memory.add), then trim every class to the new quota."""
+        for c, keep in memory.last_selection.items():
+            if task_images is not None:
+                idx = torch.as_tensor(np.ascontiguousarray(keep),
+                                      dtype=torch.int64,
+                                      device=task_images.device)
+                self._imgs[c] = task_images[idx].to(self.device)
+            else:  # fallback: one H2D upload of the selected exemplars
+                self._imgs[c] = torch.from_numpy(
+                    np.ascontiguousarray(memory._x[c])).to(self.device)
+            self._labels[c] = torch.from_numpy(
+                np.ascontiguousarray(memory._y[c]).astype(np.int64)
+            ).to(self.device)
+            self._tasks[c] = torch.from_numpy(
+                np.ascontiguousarray(memory._t[c]).astype(np.int64)
+            ).to(self.device)
+        q = memory.last_quota
+        for c in list(self._imgs):
+            self._imgs[c] = self._imgs[c][:q]
+            self._labels[c] = self._labels[c][:q]
+            self._tasks[c] = self._tasks[c][:q]
+
+    @classmethod
+    def from_memory(cls, memory, device):
+        """Rebuild after checkpoint resume (one upload, then resident)."""
+        self = cls(device)
+        for c in sorted(memory._x):
+            self._imgs[c] = torch.from_numpy(
+                np.ascontiguousarray(memory._x[c])).to(device)
+            self._labels[c] = torch.from_numpy(
+                np.ascontiguousarray(memory._y[c]).astype(np.int64)).to(device)
+            self._tasks[c] = torch.from_numpy(
+                np.ascontiguousarray(memory._t[c]).astype(np.int64)).to(device)
+        return self
+
+    def __len__(self):
+        return sum(v.shape[0] for v in self._labels.values())
+
+    def get(self):
+        """(images_u8, labels, task_ids) device tensors, memory.get() order."""
+        classes = sorted(self._imgs)
+        if not classes:
+            raise ValueError("mirror is empty")
+        return (torch.cat([self._imgs[c] for c in classes]),
+                torch.cat([self._labels[c] for c in classes]),
+                torch.cat([self._tasks[c] for c in classes]))

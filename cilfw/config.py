@@ -15,8 +15,8 @@ import argparse
 def get_args_parser():
     parser = argparse.ArgumentParser("cilfw class-incremental training", add_help=False)
 
-    # reproducibility
-    parser.add_argument("--seed", default=42, type=int)
+    # reproducibility (reference default: seed=0, template.py:16)
+    parser.add_argument("--seed", default=0, type=int)
 
     # CIL protocol (reference template.py:16-19)
     parser.add_argument("--num_bases", default=50, type=int,
@@ -36,7 +36,8 @@ def get_args_parser():
     parser.add_argument("--aa", type=str, default="rand-m9-mstd0.5-inc1",
                         help='RandAugment policy ("" disables)')
     parser.add_argument("--train_interpolation", type=str, default="bicubic")
-    parser.add_argument("--reprob", type=float, default=0.25, help="random erasing prob")
+    parser.add_argument("--reprob", type=float, default=0.0,
+                        help="random erasing prob (reference default 0.0)")
     parser.add_argument("--remode", type=str, default="pixel")
     parser.add_argument("--recount", type=int, default=1)
     parser.add_argument("--resplit", action="store_true", default=False)
@@ -73,6 +74,11 @@ def get_args_parser():
     parser.add_argument("--kd_temperature", default=2.0, type=float)
 
     # ---- cilfw-native flags (no reference counterpart) ----
+    parser.add_argument("--class_order", default=None, type=str,
+                        help="comma-separated class permutation; default: the "
+                             "reference's hardcoded CIFAR-100 order "
+                             "(template.py:201-202) for cifar100, else a "
+                             "seeded permutation")
     parser.add_argument("--dtype", default="bf16", type=str, choices=["bf16", "fp32"],
                         help="compute dtype for the backbone")
     parser.add_argument("--device", default=None, type=str,
@@ -94,6 +100,10 @@ def get_args_parser():
                         help="HBM-resident task data + on-device batch "
                              "assembly (crop/flip/normalize) — bypasses the "
                              "Python DataLoader for array-backed datasets")
+    parser.add_argument("--no_device_replay", action="store_true", default=False,
+                        help="with --gpu_data, source replay via host "
+                             "add_samples instead of the HBM-resident "
+                             "DeviceReplayMirror")
     parser.add_argument("--max_tasks", default=0, type=int,
                         help="stop after N tasks (0 = all) — e.g. the "
                              "BASELINE config[0] 2-task plumbing oracle")

@@ -5,32 +5,50 @@ consumes at CIFAR scale. When the task's images are array-backed (CIFAR /
 synthetic; not lazy ImageFolder paths), cilfw uploads the task set ONCE as
 uint8 (CIFAR-100 full train split = 150 MB — nothing against 288 GB HBM) and
 assembles every batch on-device: gather -> random crop(+pad) -> horizontal
-flip -> normalize -> bf16, all torch index ops on the GPU.
+flip -> [RandAugment / color-jitter / random-erasing, see device_augment.py]
+-> normalize -> bf16, all torch index ops on the GPU.
 
-Sharding reproduces DistributedSampler semantics exactly: same seed+epoch
-permutation on every rank, pad-by-repetition to a multiple of world size,
-rank-strided slice (cilfw/data/sampler.py contract).
+Sharding is the torch-DistributedSampler contract via the single shared
+implementation ``cilfw.data.sampler.compute_shard`` (same seed+epoch
+permutation on every rank, pad-by-repetition, rank-strided slice).
+
+``extra`` accepts already-device-resident (images_u8, labels, task_ids)
+tensors — the DeviceReplayMirror replay path (cilfw/cil/replay_gpu.py):
+exemplars are concatenated after the task tensor exactly like the
+reference's TaskSet.add_samples appends them (template.py:230-231), so the
+epoch shuffle sees an identically-ordered index space.
 """
 
 import numpy as np
 import torch
 
+from .sampler import compute_shard
+
 
 class GpuTaskLoader:
     def __init__(self, taskset, batch_size, device, mean, std, world=1, rank=0,
                  shuffle=True, seed=0, augment=True, drop_last=True,
-                 dtype=torch.bfloat16, pad=4):
+                 dtype=torch.bfloat16, pad=4, extra=None, aug_pipeline=None):
         assert taskset.x.dtype == np.uint8, \
             "GpuTaskLoader needs array-backed images (not lazy paths)"
-        self.images = torch.from_numpy(np.ascontiguousarray(taskset.x)).to(
-            device)
-        self.labels = torch.from_numpy(np.ascontiguousarray(
-            taskset.y)).to(device)
+        images = torch.from_numpy(np.ascontiguousarray(taskset.x)).to(device)
+        labels = torch.from_numpy(
+            np.ascontiguousarray(taskset.y).astype(np.int64)).to(device)
+        self.n_task = images.shape[0]
+        if extra is not None:
+            ex_imgs, ex_labels = extra[0], extra[1]
+            assert ex_imgs.dtype == torch.uint8
+            images = torch.cat([images, ex_imgs.to(device)])
+            labels = torch.cat([labels, ex_labels.to(device,
+                                                     torch.int64)])
+        self.images = images
+        self.labels = labels
         self.batch_size = batch_size
         self.device = device
         self.world, self.rank = world, rank
         self.shuffle, self.seed = shuffle, seed
         self.augment = augment
+        self.aug_pipeline = aug_pipeline  # optional DeviceAugment (RandAugment etc.)
         self.drop_last = drop_last
         self.dtype = dtype
         self.pad = pad
@@ -55,20 +73,9 @@ class GpuTaskLoader:
         return (self.num_samples + self.batch_size - 1) // self.batch_size
 
     def _shard_indices(self):
-        n = len(self.labels)
-        if self.shuffle:
-            g = torch.Generator()
-            g.manual_seed(self.seed + self.epoch)
-            idx = torch.randperm(n, generator=g)
-        else:
-            idx = torch.arange(n)
-        total = self.num_samples * self.world
-        if not self.drop_last and total > n:
-            reps = (total - n + n - 1) // n
-            idx = torch.cat([idx] + [idx] * reps)[:total]
-        else:
-            idx = idx[:total]
-        shard = idx[self.rank:total:self.world]
+        shard = compute_shard(len(self.labels), self.world, self.rank,
+                              self.shuffle, self.seed, self.epoch,
+                              self.drop_last)
         return shard.to(self.device)
 
     def __iter__(self):
@@ -78,13 +85,19 @@ class GpuTaskLoader:
         nb = len(self)
         for b in range(nb):
             idx = shard[b * self.batch_size:(b + 1) * self.batch_size]
-            imgs = self.images[idx].float()
-            if self.augment:
-                imgs = self._augment(imgs, g)
+            if self.augment and self.aug_pipeline is not None:
+                imgs = self.aug_pipeline(self.images[idx], g)
+                imgs = self._crop_flip(imgs, g)
+            else:
+                imgs = self.images[idx].float()
+                if self.augment:
+                    imgs = self._crop_flip(imgs, g)
             imgs = ((imgs - self.mean) / self.std).to(self.dtype)
+            if self.augment and self.aug_pipeline is not None:
+                imgs = self.aug_pipeline.erase(imgs, g)
             yield imgs, self.labels[idx], None
 
-    def _augment(self, imgs, g):
+    def _crop_flip(self, imgs, g):
         N, H, W, C = imgs.shape
         p = self.pad
         padded = torch.zeros(N, H + 2 * p, W + 2 * p, C, device=self.device)

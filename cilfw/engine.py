@@ -43,6 +43,17 @@ def compute_dtype(args):
     return torch.float32
 
 
+def _gpu_data_active(args, device, dataset):
+    """--gpu_data engages for array-backed datasets on cuda devices; the env
+    override CILFW_GPU_DATA_ON_CPU=1 lets tests exercise the same loader /
+    device-replay wiring on CPU (GpuTaskLoader is pure torch index ops)."""
+    import os
+    if not args.gpu_data or dataset.x.dtype != np.uint8:
+        return False
+    return (str(device).startswith("cuda")
+            or os.environ.get("CILFW_GPU_DATA_ON_CPU") == "1")
+
+
 def _to_device(inputs, targets, device, dtype):
     inputs = inputs.to(device, non_blocking=True).to(dtype)
     targets = targets.to(device, non_blocking=True)
@@ -113,9 +124,10 @@ class _GraphedStep:
     def __init__(self, step_fn, x0, y0, model):
         # torch.cuda.graph REQUIRES a side-stream warmup before capture. The
         # warmup runs fwd+bwd WITHOUT the optimizer step and with BN running
-        # stats snapshot/restored, so it leaves no trace on training state;
-        # the capture itself then EXECUTES one real step on (x0, y0) — the
-        # caller uses self.out for that batch and must not replay it again.
+        # stats snapshot/restored, so it leaves no trace on training state.
+        # Stream capture RECORDS kernels without executing them, so after
+        # capture we replay the graph once on (x0, y0) — that replay IS the
+        # capture batch's training step and fills self.out with real values.
         self.static_x = x0.clone()
         self.static_y = y0.clone()
         stats = [(b, b.clone()) for n, b in model.named_buffers()
@@ -130,6 +142,7 @@ class _GraphedStep:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self.out = step_fn(self.static_x, self.static_y)
+        self.graph.replay()  # execute the recorded step for the capture batch
 
     def __call__(self, x, y):
         self.static_x.copy_(x)
@@ -252,8 +265,7 @@ def extract_task_features(model, dataset, device, args):
     deterministic so the resulting memory is rank-identical."""
     model.eval()
     dtype = compute_dtype(args)
-    if (args.gpu_data and str(device).startswith("cuda")
-            and dataset.x.dtype == np.uint8):
+    if _gpu_data_active(args, device, dataset):
         mean, std = DATASET_STATS[getattr(args, "_stats_key", "synthetic")]
         loader = GpuTaskLoader(dataset, args.batch_size, device, mean, std,
                                shuffle=False, augment=False, drop_last=False,
@@ -285,6 +297,7 @@ def run(args):
     memory = RehearsalMemory(args.memory_size, args.herding_method,
                              args.fixed_memory, nb_total_classes=nb_classes)
     teacher = None
+    replay_mirror = None  # HBM-resident replay (DeviceReplayMirror)
     acc1s = []
     args.known_classes = 0
     start_task = 0
@@ -311,13 +324,29 @@ def run(args):
         dataset_val = scenario_val[:task_id + 1]
         args.increment_per_task = scenario_train.increments(task_id)
 
-        if task_id > 0:
-            mx, my, mt = memory.get()
-            dataset_train.add_samples(mx, my, mt)
-
         world, rank = get_world_size(), get_rank()
-        use_gpu_data = (args.gpu_data and str(device).startswith("cuda")
-                        and dataset_train.x.dtype == np.uint8)
+        use_gpu_data = _gpu_data_active(args, device, dataset_train)
+        use_device_replay = (use_gpu_data
+                             and not getattr(args, "no_device_replay", False))
+        extra = None
+        if task_id > 0:
+            if use_device_replay:
+                # replay straight from the HBM-resident mirror — exemplars
+                # never round-trip through the host (replaces the reference's
+                # memory.get() -> numpy concat, template.py:230-231). The
+                # mirror appends after the task samples exactly like
+                # add_samples, so the shuffled index space is identical.
+                if replay_mirror is None:  # resumed run: one rebuild upload
+                    from .cil.replay_gpu import DeviceReplayMirror
+                    replay_mirror = DeviceReplayMirror.from_memory(
+                        memory, device)
+                assert len(replay_mirror) == len(memory), \
+                    "device replay mirror out of sync with rehearsal memory"
+                extra = replay_mirror.get()
+            else:
+                mx, my, mt = memory.get()
+                dataset_train.add_samples(mx, my, mt)
+
         if use_gpu_data:
             mean, std = DATASET_STATS[getattr(args, "_stats_key", "synthetic")]
             train_sampler = None
@@ -325,7 +354,7 @@ def run(args):
                 dataset_train, args.batch_size, device, mean, std,
                 world=world, rank=rank, shuffle=True, seed=args.seed,
                 augment=not args.no_aug, drop_last=True,
-                dtype=compute_dtype(args))
+                dtype=compute_dtype(args), extra=extra)
             val_loader = GpuTaskLoader(
                 dataset_val, args.batch_size, device, mean, std,
                 world=world, rank=rank, shuffle=False, augment=False,
@@ -335,10 +364,13 @@ def run(args):
                                                shuffle=True, seed=args.seed)
             val_sampler = DistributedSampler(dataset_val, world, rank,
                                              shuffle=False)
+            # reference-exact: the CPU path trains on the final partial batch
+            # each epoch (template.py:236-239 has no drop_last); only the
+            # graph-captured GpuTaskLoader path needs fixed batch shapes.
             train_loader = DataLoader(
                 dataset_train, batch_size=args.batch_size,
                 sampler=train_sampler, num_workers=args.workers,
-                drop_last=True, persistent_workers=args.workers > 0)
+                drop_last=False, persistent_workers=args.workers > 0)
             val_loader = DataLoader(dataset_val, batch_size=args.batch_size,
                                     sampler=val_sampler,
                                     num_workers=args.workers)
@@ -364,6 +396,16 @@ def run(args):
         features = extract_task_features(model, dataset_train, device, args)
         rx, ry, rt = dataset_train.get_raw_samples()
         memory.add(rx, ry, rt, features)
+        if use_device_replay:
+            # gather the just-selected exemplars from the already-resident
+            # task tensor (device-side; zero host traffic) and trim quotas
+            if replay_mirror is None:
+                from .cil.replay_gpu import DeviceReplayMirror
+                replay_mirror = DeviceReplayMirror(device)
+            replay_mirror.update(
+                memory,
+                task_images=train_loader.images[:train_loader.n_task],
+                task_id=task_id)
 
         engine.detach()
         args.known_classes += args.increment_per_task  # before snapshot: the

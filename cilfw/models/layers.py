@@ -44,6 +44,14 @@ class BatchNormAct2d(nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
 
     def forward(self, x, residual=None):
+        # A frozen teacher's (mean, invstd) are precomputed once
+        # (CilModel.cast_compute_weights_); that is only valid while the
+        # running stats never change — a train-mode forward would update them
+        # and silently desynchronize the cached pair.
+        assert not (self.training
+                    and getattr(self.running_mean, "_cilfw_frozen", None)
+                    is not None), \
+            "BN has precomputed frozen eval stats but is in train mode"
         if residual is not None:
             # fused y = relu(bn(x) + residual) — residual-block tail
             return CF.batchnorm_add_relu(x, residual, self.weight, self.bias,

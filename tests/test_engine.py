@@ -39,6 +39,18 @@ def test_determinism_same_seed():
 
 
 @pytest.mark.timeout(900)
+def test_device_replay_matches_host_replay(monkeypatch):
+    """Engine path: the HBM-resident DeviceReplayMirror replay source must
+    produce the same trajectory as the host add_samples path (identical index
+    space + shuffle + augmentation stream), here exercised on CPU via the
+    CILFW_GPU_DATA_ON_CPU override."""
+    monkeypatch.setenv("CILFW_GPU_DATA_ON_CPU", "1")
+    host = run(_args(["--gpu_data", "--no_device_replay"], epochs=2))
+    dev = run(_args(["--gpu_data"], epochs=2))
+    assert host == dev
+
+
+@pytest.mark.timeout(900)
 def test_checkpoint_resume_continues(tmp_path):
     base = ["--output_dir", str(tmp_path)]
     full = run(_args(base, epochs=2))

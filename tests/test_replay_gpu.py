@@ -46,3 +46,88 @@ def test_device_store_gpu():
     imgs, labels = st.sample(32)
     assert imgs.is_cuda and imgs.dtype == torch.bfloat16
     assert labels.is_cuda
+
+
+# ---- DeviceReplayMirror: the engine's default --gpu_data replay source ----
+
+from cilfw.cil.replay_gpu import DeviceReplayMirror  # noqa: E402
+
+
+def _task(rng, classes, n_per_class, task_id, hw=8):
+    x = rng.integers(0, 255, (len(classes) * n_per_class, hw, hw, 3),
+                     dtype=np.uint8)
+    y = np.repeat(classes, n_per_class)
+    t = np.full(len(y), task_id, dtype=np.int64)
+    return x, y, t
+
+
+def test_mirror_matches_memory_over_tasks():
+    """Two-task flow: the device mirror's content must stay byte-identical to
+    the host RehearsalMemory (same herding picks, same quota shrink, same
+    sorted-class concat order)."""
+    mem = RehearsalMemory(memory_size=30)
+    mirror = DeviceReplayMirror("cpu")
+    rng = np.random.default_rng(1)
+
+    for task_id, classes in enumerate([[0, 1, 2], [3, 4]]):
+        x, y, t = _task(rng, classes, 20, task_id)
+        feats = torch.from_numpy(
+            rng.standard_normal((len(y), 16)).astype(np.float32))
+        mem.add(x, y, t, feats)
+        task_images_dev = torch.from_numpy(x)  # "already uploaded" tensor
+        mirror.update(mem, task_images=task_images_dev, task_id=task_id)
+
+        mx, my, mt = mem.get()
+        gx, gy, gt = mirror.get()
+        assert len(mirror) == len(mem)
+        np.testing.assert_array_equal(gx.numpy(), mx)
+        np.testing.assert_array_equal(gy.numpy(), my)
+        np.testing.assert_array_equal(gt.numpy(), mt)
+
+
+def test_mirror_from_memory_rebuild():
+    """Resume path: rebuilding from host memory gives identical content."""
+    mem = RehearsalMemory(memory_size=24)
+    rng = np.random.default_rng(2)
+    x, y, t = _task(rng, [0, 1], 15, 0)
+    mem.add(x, y, t, torch.from_numpy(
+        rng.standard_normal((len(y), 8)).astype(np.float32)))
+    mirror = DeviceReplayMirror.from_memory(mem, "cpu")
+    mx, my, mt = mem.get()
+    gx, gy, gt = mirror.get()
+    np.testing.assert_array_equal(gx.numpy(), mx)
+    np.testing.assert_array_equal(gy.numpy(), my)
+    np.testing.assert_array_equal(gt.numpy(), mt)
+
+
+def test_loader_extra_equals_add_samples():
+    """GpuTaskLoader with device-resident `extra` replay must produce
+    bit-identical batches to the host add_samples path (same index space,
+    same shuffle, same augmentation stream)."""
+    from cilfw.data.gpu_pipeline import GpuTaskLoader
+    from cilfw.data.scenario import TaskSet
+
+    rng = np.random.default_rng(3)
+    tx, ty, tt = _task(rng, [5, 6], 30, 1)
+    rx, ry, rt = _task(rng, [0, 1], 10, 0)
+
+    host_task = TaskSet(tx.copy(), ty.copy(), tt.copy())
+    host_task.add_samples(rx, ry, rt)
+    mean, std = (0.5, 0.5, 0.5), (0.25, 0.25, 0.25)
+    kw = dict(batch_size=16, device="cpu", mean=mean, std=std, world=2,
+              rank=1, shuffle=True, seed=7, augment=True, drop_last=True,
+              dtype=torch.float32)
+    loader_host = GpuTaskLoader(host_task, **kw)
+
+    dev_task = TaskSet(tx.copy(), ty.copy(), tt.copy())
+    extra = (torch.from_numpy(rx), torch.from_numpy(ry.astype(np.int64)),
+             torch.from_numpy(rt))
+    loader_dev = GpuTaskLoader(dev_task, extra=extra, **kw)
+
+    assert len(loader_host) == len(loader_dev)
+    for ep in range(2):
+        loader_host.set_epoch(ep)
+        loader_dev.set_epoch(ep)
+        for (ia, la, _), (ib, lb, _) in zip(loader_host, loader_dev):
+            assert torch.equal(ia, ib)
+            assert torch.equal(la, lb)
