@@ -73,7 +73,17 @@ void reduce_slabs_f32_kernel(const float* __restrict__ ws,
 
 // ============================== forward ==============================
 
-template <int BKT, int BMX = BM, int BNX = BN>
+// Branchless OOB handling: gather loads use a CLAMPED (always-valid) address
+// and the loaded value is zeroed by cndmask when out of bounds. The
+// conditional-load form compiles to exec-mask branches around every load
+// (63 s_and_saveexec waterfalls in the round-1 K-loop, 1758-instruction body
+// for 8 MFMAs); the select form keeps the loop straight-line.
+__device__ __forceinline__ int4 masked_i4(int4 v, bool ok) {
+  int4 z{0, 0, 0, 0};
+  return ok ? v : z;
+}
+
+template <int BKT, int BMX = BM, int BNX = BN, bool FAST = true>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
                        const bf16_t* __restrict__ w,
@@ -102,13 +112,11 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   for (int rr = 0; rr < NR; ++rr) {
     int m = m0 + arow + rr * 128;
     arow_ok[rr] = m < M;
-    int n = 0, aho = 0, awo = 0;
-    if (arow_ok[rr]) {
-      n = m / (g.Ho * g.Wo);
-      int rem = m - n * (g.Ho * g.Wo);
-      aho = rem / g.Wo;
-      awo = rem - aho * g.Wo;
-    }
+    int mc = min(m, M - 1);  // clamped: address math always valid
+    int n = mc / (g.Ho * g.Wo);
+    int rem = mc - n * (g.Ho * g.Wo);
+    int aho = rem / g.Wo;
+    int awo = rem - aho * g.Wo;
     an[rr] = n;
     ahb[rr] = aho * g.stride - g.pad;
     awb[rr] = awo * g.stride - g.pad2;
@@ -126,7 +134,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
   // incremental im2col decomposition (fast path): per q-chunk (r, s, c0)
   // advance by compare/sub each K-step instead of div/mod every call
   int inc_r[NQ], inc_s[NQ], inc_c0[NQ];
-  if (fast_a) {
+  if (FAST) {
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
       int k = kt0 * BKT + q * 32 + ahalf * 16;
@@ -142,20 +150,19 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
       // A chunk: 16 elems at col q*32 + ahalf*16 (within one (r,s): C%16==0)
-      if (fast_a) {
+      if (FAST) {
         int c0 = inc_c0[q], r = inc_r[q], s = inc_s[q];
 #pragma unroll
         for (int rr = 0; rr < NR; ++rr) {
           int hi = ahb[rr] + r, wi = awb[rr] + s;
-          if (arow_ok[rr] && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W) {
-            const int4* src = (const int4*)&x[(((long)an[rr] * g.H + hi)
-                                               * g.W + wi) * g.C + c0];
-            areg[rr][2 * q] = src[0];
-            areg[rr][2 * q + 1] = src[1];
-          } else {
-            areg[rr][2 * q] = int4{0, 0, 0, 0};
-            areg[rr][2 * q + 1] = int4{0, 0, 0, 0};
-          }
+          bool ok = arow_ok[rr] & ((unsigned)hi < (unsigned)g.H)
+                    & ((unsigned)wi < (unsigned)g.W);
+          int hic = min(max(hi, 0), g.H - 1);
+          int wic = min(max(wi, 0), g.W - 1);
+          const int4* src = (const int4*)&x[(((long)an[rr] * g.H + hic)
+                                             * g.W + wic) * g.C + c0];
+          areg[rr][2 * q] = masked_i4(src[0], ok);
+          areg[rr][2 * q + 1] = masked_i4(src[1], ok);
         }
         c0 += BKT;
         while (c0 >= g.C) {
@@ -192,10 +199,13 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
         const int bn = (t & 63) + h * 64;
         const int bk8 = ((t >> 6) & 3) * 8 + q * 32;
         const bool n_ok = ko0 + bn < g.K;
+        const int bcol = min(ko0 + bn, g.K - 1);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           int k = k0 + bk8 + j;
-          breg[q][h][j] = (n_ok && k < CRS) ? w[(long)k * g.K + ko0 + bn] : 0;
+          bool ok = n_ok & (k < CRS);
+          bf16_t v = w[(long)min(k, CRS - 1) * g.K + bcol];
+          breg[q][h][j] = ok ? v : (bf16_t)0;
         }
       }
     }
@@ -281,7 +291,7 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 // ============================== backward data ==============================
 // dX[M=N*H*W, C] = gather(dY)[M, R*S*K] @ B where B[(r,s,k)][c] = W[r,s,c,k].
 
-template <int BKT>
+template <int BKT, bool FAST = true>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
                             const bf16_t* __restrict__ w,
@@ -302,14 +312,12 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
 
   const int arow = t >> 1, ahalf = t & 1;
   int m = m0 + arow;
-  int an = 0, ahi = 0, awi = 0;
   bool arow_ok = m < M;
-  if (arow_ok) {
-    an = m / (g.H * g.W);
-    int rem = m - an * (g.H * g.W);
-    ahi = rem / g.W;
-    awi = rem - ahi * g.W;
-  }
+  int mc = min(m, M - 1);
+  int an = mc / (g.H * g.W);
+  int mrem = mc - an * (g.H * g.W);
+  int ahi = mrem / g.W;
+  int awi = mrem - ahi * g.W;
 
   f32x4 acc[4][2];
 #pragma unroll
@@ -321,7 +329,9 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
   __align__(16) bf16_t breg[NQ][8];
 
   int inc_r[NQ], inc_s[NQ], inc_kc[NQ];
-  if (fast_a) {
+  // B-gather incremental state (c row fixed = t>>2; k advances by BKT/step)
+  int binc_rs[NQ], binc_kc[NQ];
+  if (FAST) {
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
       int k = kt0 * BKT + q * 32 + ahalf * 16;
@@ -329,6 +339,9 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       inc_kc[q] = k - rs * g.K;
       inc_r[q] = rs / g.S;
       inc_s[q] = rs - inc_r[q] * g.S;
+      int bk = kt0 * BKT + (t & 3) * 8 + q * 32;
+      binc_rs[q] = bk / g.K;
+      binc_kc[q] = bk - binc_rs[q] * g.K;
     }
   }
 
@@ -336,22 +349,19 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
     const int k0 = kt * BKT;
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
-      if (fast_a) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
+      if (FAST) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
         int kc0 = inc_kc[q], r = inc_r[q], s = inc_s[q];
         int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
-        bool ok = arow_ok && ho2 >= 0 && wo2 >= 0 &&
-                  (ho2 % g.stride) == 0 && (wo2 % g.stride) == 0;
         int ho = ho2 / g.stride, wo = wo2 / g.stride;
-        ok = ok && ho < g.Ho && wo < g.Wo;
-        if (ok) {
-          const int4* src = (const int4*)&dy[(((long)an * g.Ho + ho) * g.Wo
-                                              + wo) * g.K + kc0];
-          areg[2 * q] = src[0];
-          areg[2 * q + 1] = src[1];
-        } else {
-          areg[2 * q] = int4{0, 0, 0, 0};
-          areg[2 * q + 1] = int4{0, 0, 0, 0};
-        }
+        bool ok = arow_ok & (ho2 >= 0) & (wo2 >= 0) &
+                  (ho2 - ho * g.stride == 0) & (wo2 - wo * g.stride == 0) &
+                  (ho < g.Ho) & (wo < g.Wo);
+        int hoc = min(max(ho, 0), g.Ho - 1);
+        int woc = min(max(wo, 0), g.Wo - 1);
+        const int4* src = (const int4*)&dy[(((long)an * g.Ho + hoc) * g.Wo
+                                            + woc) * g.K + kc0];
+        areg[2 * q] = masked_i4(src[0], ok);
+        areg[2 * q + 1] = masked_i4(src[1], ok);
         kc0 += BKT;
         while (kc0 >= g.K) {
           kc0 -= g.K;
@@ -383,7 +393,17 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       // B: thread c = t>>2 (0..63), slot = t&3 -> 8 contiguous kc (K % 8 == 0)
       const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
       int k = k0 + bkk;
-      if (k < RSK && c0 + bc < g.C) {
+      const int bcc = min(c0 + bc, g.C - 1);
+      if (FAST) {
+        int rs = binc_rs[q], kc = binc_kc[q];
+        bool ok = (k < RSK) & (c0 + bc < g.C);
+        int rsc = min(rs, g.R * g.S - 1);
+        const int4* src = (const int4*)&w[((long)rsc * g.C + bcc) * g.K + kc];
+        *(int4*)breg[q] = masked_i4(*src, ok);
+        kc += BKT;
+        while (kc >= g.K) { kc -= g.K; ++rs; }
+        binc_rs[q] = rs; binc_kc[q] = kc;
+      } else if (k < RSK && c0 + bc < g.C) {
         int rs = k / g.K, kc = k - rs * g.K;
         *(int4*)breg[q] = *(const int4*)&w[((long)rs * g.C + c0 + bc) * g.K
                                            + kc];
@@ -493,14 +513,12 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
 
   const int arow = t >> 1, ahalf = t & 1;
   int m = m0 + arow;
-  int an = 0, ah2 = 0, aw2 = 0;
   bool arow_ok = m < Mc;
-  if (arow_ok) {
-    an = m / (H2 * W2);
-    int rem = m - an * (H2 * W2);
-    ah2 = rem / W2;
-    aw2 = rem - ah2 * W2;
-  }
+  int mc_ = min(m, Mc - 1);
+  int an = mc_ / (H2 * W2);
+  int mrem = mc_ - an * (H2 * W2);
+  int ah2 = mrem / W2;
+  int aw2 = mrem - ah2 * W2;
 
   f32x4 acc[4][2];
 #pragma unroll
@@ -511,8 +529,9 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
   int4 areg[2 * NQ];
   __align__(16) bf16_t breg[NQ][8];
 
-  // incremental (jr, js, kc) decomposition per q-chunk
+  // incremental (jr, js, kc) decomposition per q-chunk (A and B gathers)
   int inc_jr[NQ], inc_js[NQ], inc_kc[NQ];
+  int binc_jr[NQ], binc_js[NQ], binc_kc[NQ];
 #pragma unroll
   for (int q = 0; q < NQ; ++q) {
     int k = q * 32 + ahalf * 16;  // kt0 == 0 (no ksplit)
@@ -520,24 +539,27 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
     inc_kc[q] = k - rs * g.K;
     inc_jr[q] = rs / ns;
     inc_js[q] = rs - inc_jr[q] * ns;
+    int bk = (t & 3) * 8 + q * 32;
+    int brs = bk / g.K;
+    binc_kc[q] = bk - brs * g.K;
+    binc_jr[q] = brs / ns;
+    binc_js[q] = brs - binc_jr[q] * ns;
   }
 
   auto stage_to_regs = [&](int kt) {
-    const int k0 = kt * BKT;
 #pragma unroll
     for (int q = 0; q < NQ; ++q) {
       int kc0 = inc_kc[q], jr = inc_jr[q], js = inc_js[q];
       {
         int ho = ah2 + padh - jr, wo = aw2 + padw - js;
-        if (arow_ok && ho >= 0 && ho < g.Ho && wo >= 0 && wo < g.Wo) {
-          const int4* src = (const int4*)&dy[(((long)an * g.Ho + ho) * g.Wo
-                                              + wo) * g.K + kc0];
-          areg[2 * q] = src[0];
-          areg[2 * q + 1] = src[1];
-        } else {
-          areg[2 * q] = int4{0, 0, 0, 0};
-          areg[2 * q + 1] = int4{0, 0, 0, 0};
-        }
+        bool ok = arow_ok & ((unsigned)ho < (unsigned)g.Ho)
+                  & ((unsigned)wo < (unsigned)g.Wo);
+        int hoc = min(max(ho, 0), g.Ho - 1);
+        int woc = min(max(wo, 0), g.Wo - 1);
+        const int4* src = (const int4*)&dy[(((long)an * g.Ho + hoc) * g.Wo
+                                            + woc) * g.K + kc0];
+        areg[2 * q] = masked_i4(src[0], ok);
+        areg[2 * q + 1] = masked_i4(src[1], ok);
         kc0 += BKT;
         while (kc0 >= g.K) {
           kc0 -= g.K;
@@ -547,15 +569,22 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
       }
       // B: Bs[c][kk] = w[((r0+2jr)*S + s0+2js)*C + c][kc]
       const int bc = t >> 2, bkk = (t & 3) * 8 + q * 32;
-      int k = k0 + bkk;
-      if (k < RSK && c0 + bc < g.C) {
-        int rs = k / g.K, kc = k - rs * g.K;
-        int jr2 = rs / ns, js2 = rs - jr2 * ns;
-        int rs_orig = (r0 + 2 * jr2) * g.S + s0 + 2 * js2;
-        *(int4*)breg[q] = *(const int4*)&w[((long)rs_orig * g.C + c0 + bc)
-                                           * g.K + kc];
-      } else {
-        *(int4*)breg[q] = int4{0, 0, 0, 0};
+      int k = kt * BKT + bkk;
+      {
+        int kc = binc_kc[q], jr2 = binc_jr[q], js2 = binc_js[q];
+        bool ok = (k < RSK) & (c0 + bc < g.C);
+        int bcc = min(c0 + bc, g.C - 1);
+        int jr2c = min(jr2, nr - 1), js2c = min(js2, ns - 1);
+        int rs_orig = (r0 + 2 * jr2c) * g.S + s0 + 2 * js2c;
+        const int4* src = (const int4*)&w[((long)rs_orig * g.C + bcc)
+                                          * g.K + kc];
+        *(int4*)breg[q] = masked_i4(*src, ok);
+        kc += BKT;
+        while (kc >= g.K) {
+          kc -= g.K;
+          if (++js2 == ns) { js2 = 0; ++jr2; }
+        }
+        binc_kc[q] = kc; binc_jr[q] = jr2; binc_js[q] = js2;
       }
     }
   };
@@ -635,6 +664,7 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
 #define WBS_OFF(buf) (2 * WBM * WLP + (buf) * BN * WLP)
 #define WLDS_ELEMS (2 * WBM * WLP + 2 * BN * WLP)
 
+template <bool FAST>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
                               const bf16_t* __restrict__ x,
@@ -649,15 +679,15 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   const int wave = t >> 6, wr = wave >> 1, wc = wave & 1;
 
   const int amm = t >> 3, agrp = t & 7;
-  int r_ = 0, s_ = 0, cbase_ = 0;
   const int rowb = rs0 + agrp * 8;
   bool agrp_ok = rowb < CRS;
-  if (agrp_ok) {
-    int rs = rowb / g.C;
-    cbase_ = rowb - rs * g.C;
-    r_ = rs / g.S;
-    s_ = rs - r_ * g.S;
-  }
+  // clamp to the last 8-aligned chunk (FAST: C%8==0 => CRS%8==0), so the
+  // int4 gather of a clamped row stays 8-aligned and inside the pixel row
+  const int rowbc = min(rowb, max(CRS - 8, 0));
+  int rs_ = rowbc / g.C;
+  int cbase_ = rowbc - rs_ * g.C;
+  int r_ = rs_ / g.S;
+  int s_ = rs_ - r_ * g.S;
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -671,24 +701,26 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
 
   const bool flat = (g.H == 1 && g.W == 1 && g.Ho == 1 && g.Wo == 1);
   auto gather_one = [&](int m, bf16_t* adst, bf16_t* bdst) {
-    int n = 0, ho = 0, wo = 0;
     const bool m_ok = m < me;
-    if (m_ok && !flat) {
-      n = m / HoWo;
-      int rem = m - n * HoWo;
+    const int mcl = min(m, me - 1);
+    int n, ho = 0, wo = 0;
+    if (!flat) {  // wave-uniform condition
+      n = mcl / HoWo;
+      int rem = mcl - n * HoWo;
       ho = rem / g.Wo;
       wo = rem - ho * g.Wo;
-    } else if (m_ok) {
-      n = m;  // flat im2col geometry: one "pixel" per row
+    } else {
+      n = mcl;  // flat im2col geometry: one "pixel" per row
     }
-    if (fast_a) {
+    if (FAST) {
       int hi = ho * g.stride - g.pad + r_;
       int wi = wo * g.stride - g.pad2 + s_;
-      if (agrp_ok && m_ok && hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-        *(int4*)adst = *(const int4*)&x[(((long)n * g.H + hi) * g.W + wi)
-                                        * g.C + cbase_];
-      else
-        *(int4*)adst = int4{0, 0, 0, 0};
+      bool ok = agrp_ok & m_ok & ((unsigned)hi < (unsigned)g.H)
+                & ((unsigned)wi < (unsigned)g.W);
+      int hic = min(max(hi, 0), g.H - 1);
+      int wic = min(max(wi, 0), g.W - 1);
+      *(int4*)adst = masked_i4(*(const int4*)&x[(((long)n * g.H + hic) * g.W
+                                                 + wic) * g.C + cbase_], ok);
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -705,8 +737,9 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
         adst[j] = v;
       }
     }
-    if (m_ok && ko0 + agrp * 8 + 8 <= g.K) {
-      *(int4*)bdst = *(const int4*)&dy[(long)m * g.K + ko0 + agrp * 8];
+    if (ko0 + BN <= g.K) {  // block-uniform: full 64-col dY tile
+      *(int4*)bdst = masked_i4(
+          *(const int4*)&dy[(long)mcl * g.K + ko0 + agrp * 8], m_ok);
     } else if (m_ok) {
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -897,21 +930,26 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
               (cdiv(M, BM) * cdiv(K, BN) * ksplit < 768);
   int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, bm), cdiv(K, BN), ksplit);
+#define LAUNCH_FWD(BKT_, BM_)                                                 \
+  do {                                                                        \
+    if (fast_a)                                                               \
+      hipLaunchKernelGGL((conv2d_fwd_kernel<BKT_, BM_, BN, true>), grid,      \
+                         dim3(NTHREADS), 0, (hipStream_t)stream,              \
+                         (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,      \
+                         (float*)ws, g, M, CRS, nk, fast_a, ksplit);          \
+    else                                                                      \
+      hipLaunchKernelGGL((conv2d_fwd_kernel<BKT_, BM_, BN, false>), grid,     \
+                         dim3(NTHREADS), 0, (hipStream_t)stream,              \
+                         (const bf16_t*)x, (const bf16_t*)w, (bf16_t*)y,      \
+                         (float*)ws, g, M, CRS, nk, fast_a, ksplit);          \
+  } while (0)
   if (bm == 256)
-    hipLaunchKernelGGL((conv2d_fwd_kernel<32, 256>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)x,
-                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
-                       nk, fast_a, ksplit);
+    LAUNCH_FWD(32, 256);
   else if (use64)
-    hipLaunchKernelGGL((conv2d_fwd_kernel<64>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)x,
-                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
-                       nk, fast_a, ksplit);
+    LAUNCH_FWD(64, 128);
   else
-    hipLaunchKernelGGL((conv2d_fwd_kernel<32>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)x,
-                       (const bf16_t*)w, (bf16_t*)y, (float*)ws, g, M, CRS,
-                       nk, fast_a, ksplit);
+    LAUNCH_FWD(32, 128);
+#undef LAUNCH_FWD
   if (ksplit > 1) {
     long len = (long)M * K;
     hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -955,16 +993,26 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
-  if (use64)
-    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)dy,
-                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
-                       nk, fast_a, ksplit);
+  if (use64 && fast_a)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64, true>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  else if (use64)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64, false>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  else if (fast_a)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32, true>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
   else
-    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)dy,
-                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
-                       nk, fast_a, ksplit);
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32, false>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
   if (ksplit > 1) {
     long len = (long)M * C;
     hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -987,16 +1035,26 @@ void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
-  if (use64)
-    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)dy,
-                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
-                       nk, fast_a, ksplit);
+  if (use64 && fast_a)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64, true>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  else if (use64)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<64, false>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
+  else if (fast_a)
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32, true>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
   else
-    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32>), grid, dim3(NTHREADS), 0,
-                       (hipStream_t)stream, (const bf16_t*)dy,
-                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws, g, M, RSK,
-                       nk, fast_a, ksplit);
+    hipLaunchKernelGGL((conv2d_bwd_data_kernel<32, false>), grid,
+                       dim3(NTHREADS), 0, (hipStream_t)stream,
+                       (const bf16_t*)dy, (const bf16_t*)w, (bf16_t*)dx,
+                       (float*)ws, g, M, RSK, nk, fast_a, ksplit);
   if (ksplit > 1) {
     long len = (long)M * C;
     hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -1037,9 +1095,16 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
   slice_len = cdiv(slice_len, WBK) * WBK;
   int fast_a = (C % 8 == 0);
   dim3 grid(cdiv(CRS, WBM), cdiv(K, BN), nslices);
-  hipLaunchKernelGGL(conv2d_bwd_weight_kernel, grid, dim3(NTHREADS), 0,
-                     (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)x,
-                     (float*)ws, g, M, CRS, slice_len, fast_a);
+  if (fast_a)
+    hipLaunchKernelGGL(conv2d_bwd_weight_kernel<true>, grid, dim3(NTHREADS),
+                       0, (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)x, (float*)ws, g, M, CRS, slice_len,
+                       fast_a);
+  else
+    hipLaunchKernelGGL(conv2d_bwd_weight_kernel<false>, grid, dim3(NTHREADS),
+                       0, (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)x, (float*)ws, g, M, CRS, slice_len,
+                       fast_a);
   long len = (long)CRS * K;
   hipLaunchKernelGGL(reduce_slabs_f32_kernel,
                      dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
