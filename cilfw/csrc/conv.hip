@@ -659,10 +659,38 @@ void conv2d_bwd_data_s2_kernel(const bf16_t* __restrict__ dy,
 
 #define WBM 64
 #define WBK 64
-#define WLP (WBK + 8)
-#define WAS_OFF(buf) ((buf) * WBM * WLP)
-#define WBS_OFF(buf) (2 * WBM * WLP + (buf) * BN * WLP)
-#define WLDS_ELEMS (2 * WBM * WLP + 2 * BN * WLP)
+// LDS images are [rowgrp(4)][m(64)][16] per operand — m-major within a
+// 16-row column group (32 B m-stride, the attention-V tr-read shape): the
+// gather's natural int4 (8 rows of one m) lands as ONE b128 write, and the
+// MFMA fragment (one row, 8 m's) comes back as TWO ds_read_b64_tr_b16.
+// Replaces the round-1 [row][m] image whose transpose staging was 32 scalar
+// ds_write_b16 per thread per K-step (the biggest measured kernel cost).
+// +16-elem pad between 16-col groups: staging's 8 write lanes then start at
+// 8 distinct 4-dword bank ranges (conflict-free ds_write_b128)
+#define WGRP_ELEMS (WBK * 16 + 16)
+#define WIMG_ELEMS (4 * WGRP_ELEMS)      // one operand image (64 rows)
+#define WAS_OFF(buf) ((buf) * WIMG_ELEMS)
+#define WBS_OFF(buf) (2 * WIMG_ELEMS + (buf) * WIMG_ELEMS)
+#define WLDS_ELEMS (4 * WIMG_ELEMS)
+
+// two transpose-reads -> one MFMA operand fragment.
+// Measured gfx950 semantics (tools/probe/trprobe.hip): each lane loads 8
+// ALIGNED bytes at its own address; per 16-lane group, "row" j is the 32 B
+// loaded by lanes 4j..4j+3 and lane s receives element s of each row. So
+// with lane address = img + (mbase + ((l>>2)&3))*32B + (l&3)*8B over a
+// [m][16 col] image, lane s = l&15 gets col s at m = mbase+j — one MFMA
+// operand column slice; the offset:128 twin covers m = mbase+4..7.
+DEV bf16x8 tr_frag(unsigned lds_byte) {
+  typedef __attribute__((ext_vector_type(4))) short s4;
+  s4 lo, hi;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(lo), "=&v"(hi)
+      : "v"(lds_byte));
+  return __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+}
 
 template <bool FAST>
 __global__ __launch_bounds__(NTHREADS)
@@ -756,19 +784,15 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
   };
 
   auto regs_to_lds = [&](int buf) {
-    bf16_t* As = &lds[WAS_OFF(buf)];
-    bf16_t* Bs = &lds[WBS_OFF(buf)];
-    // column XOR-swizzle by (row>>3)&7 decorrelates the 8-row write stride
-    // (8*WLP/2 = 0 mod 32 banks -> 16-way conflict without it); reads apply
-    // the same XOR (kb is 8-aligned, so chunks stay contiguous b128s).
-    const int sw = (agrp & 7) << 3;
+    // one b128 write per gather: image[rowgrp][m][16], rowgrp = agrp>>1,
+    // in-group column base = (agrp&1)*8
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        As[(agrp * 8 + j) * WLP + ((amm + h * 32) ^ sw)] = areg[h][j];
-        Bs[(agrp * 8 + j) * WLP + ((amm + h * 32) ^ sw)] = breg[h][j];
-      }
+    for (int h = 0; h < 2; ++h) {
+      const int e = (agrp >> 1) * WGRP_ELEMS + (amm + h * 32) * 16
+                    + (agrp & 1) * 8;
+      *(int4*)&lds[WAS_OFF(buf) + e] = *(int4*)areg[h];
+      *(int4*)&lds[WBS_OFF(buf) + e] = *(int4*)breg[h];
+    }
   };
 
   const int nk = cdiv_i(me - ms, WBK);
@@ -777,30 +801,36 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     regs_to_lds(0);
   }
   __syncthreads();
+  FragIdx fi = frag_idx();
+  // per-lane tr-read byte bases: lane covers m-row (fi.quad*8 + ((l>>2)&3)),
+  // byte slot (l&3)*8 of the 32 B row; received = col (l&15), m quad*8+j
+  const unsigned lds0 = (unsigned)(uintptr_t)&lds[0];
+  const int lane = t & 63;
+  const unsigned lane_e = (unsigned)((fi.quad * 8 + ((lane >> 2) & 3)) * 16
+                                     + (lane & 3) * 4);
   for (int kt = 0; kt < nk; ++kt) {
     int cur = kt & 1;
     if (kt + 1 < nk) stage_to_regs(ms + (kt + 1) * WBK);
     {
-      FragIdx fi = frag_idx();
-      const bf16_t* As = &lds[WAS_OFF(cur)];
-      const bf16_t* Bs = &lds[WBS_OFF(cur)];
 #pragma unroll
       for (int kh = 0; kh < 2; ++kh) {
-        const int kb = fi.quad * 8 + kh * 32;
-#pragma unroll
-        for (int mr = 0; mr < 2; ++mr) {
-          int row = wr * 32 + mr * 16 + fi.half;
-          bf16x8 a = *(const bf16x8*)&As[row * WLP
-                                         + (kb ^ (((row >> 3) & 7) << 3))];
-#pragma unroll
-          for (int nr = 0; nr < 2; ++nr) {
-            int col = wc * 32 + nr * 16 + fi.half;
-            bf16x8 b = *(const bf16x8*)&Bs[col * WLP
-                                           + (kb ^ (((col >> 3) & 7) << 3))];
-            acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a, b, acc[mr][nr], 0, 0, 0);
-          }
-        }
+        const unsigned khe = lane_e + kh * 32 * 16;
+        bf16x8 a0 = tr_frag(lds0 + 2 * (WAS_OFF(cur)
+                            + (wr * 2 + 0) * WGRP_ELEMS + khe));
+        bf16x8 a1 = tr_frag(lds0 + 2 * (WAS_OFF(cur)
+                            + (wr * 2 + 1) * WGRP_ELEMS + khe));
+        bf16x8 b0 = tr_frag(lds0 + 2 * (WBS_OFF(cur)
+                            + (wc * 2 + 0) * WGRP_ELEMS + khe));
+        bf16x8 b1 = tr_frag(lds0 + 2 * (WBS_OFF(cur)
+                            + (wc * 2 + 1) * WGRP_ELEMS + khe));
+        acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0,
+                                                            acc[0][0], 0, 0, 0);
+        acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1,
+                                                            acc[0][1], 0, 0, 0);
+        acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0,
+                                                            acc[1][0], 0, 0, 0);
+        acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1,
+                                                            acc[1][1], 0, 0, 0);
       }
     }
     if (kt + 1 < nk) {  // single barrier per K-step (see fwd kernel note)
@@ -809,7 +839,6 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     }
   }
 
-  FragIdx fi = frag_idx();
   const long slab = (long)blockIdx.z * CRS * g.K;
 #pragma unroll
   for (int mr = 0; mr < 2; ++mr)
