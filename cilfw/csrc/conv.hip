@@ -24,7 +24,20 @@
 
 struct ConvGeom {
   int N, H, W, C, K, R, S, stride, pad, pad2, Ho, Wo;  // pad2 = w-axis pad
+  // round-up magic multipliers for division by Ho*Wo and Wo in per-K-step
+  // gathers (exact for n < 2^22, divisor < 2^14 — launcher asserts):
+  // q = (n * m) >> 40 == n / d. Runtime-divisor division otherwise expands
+  // to a ~40-instruction v_rcp sequence inside the hot loop.
+  unsigned long long m_howo, m_wo;
 };
+
+static inline unsigned long long magic40(int d) {
+  return d > 0 ? ((1ULL << 40) / (unsigned)d + 1) : 0;
+}
+
+DEV unsigned mdiv40(unsigned n, unsigned long long m) {
+  return (unsigned)(((unsigned long long)n * m) >> 40);
+}
 
 // sum ksplit fp32 slabs [ns][len] -> bf16 out[len]
 __global__ __launch_bounds__(NTHREADS)
@@ -352,9 +365,13 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
       if (FAST) {  // K % 16 == 0: a 16-chunk stays inside one (r,s)
         int kc0 = inc_kc[q], r = inc_r[q], s = inc_s[q];
         int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
-        int ho = ho2 / g.stride, wo = wo2 / g.stride;
-        bool ok = arow_ok & (ho2 >= 0) & (wo2 >= 0) &
-                  (ho2 - ho * g.stride == 0) & (wo2 - wo * g.stride == 0) &
+        // stride is 1 or 2 everywhere cilfw launches this: a shift-select
+        // replaces the runtime division (a ~40-instruction expansion here)
+        const bool s2d = g.stride == 2;
+        int ho = s2d ? (ho2 >> 1) : ho2;
+        int wo = s2d ? (wo2 >> 1) : wo2;
+        bool par = !s2d | (((ho2 | wo2) & 1) == 0);
+        bool ok = arow_ok & (ho2 >= 0) & (wo2 >= 0) & par &
                   (ho < g.Ho) & (wo < g.Wo);
         int hoc = min(max(ho, 0), g.Ho - 1);
         int woc = min(max(wo, 0), g.Wo - 1);
@@ -732,10 +749,10 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
     const bool m_ok = m < me;
     const int mcl = min(m, me - 1);
     int n, ho = 0, wo = 0;
-    if (!flat) {  // wave-uniform condition
-      n = mcl / HoWo;
+    if (!flat) {  // wave-uniform condition; magic division (hot loop)
+      n = (int)mdiv40((unsigned)mcl, g.m_howo);
       int rem = mcl - n * HoWo;
-      ho = rem / g.Wo;
+      ho = (int)mdiv40((unsigned)rem, g.m_wo);
       wo = rem - ho * g.Wo;
     } else {
       n = mcl;  // flat im2col geometry: one "pixel" per row
@@ -910,6 +927,7 @@ void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
 // ============================== launchers ==============================
 
 #include <stdlib.h>
+#include <stdio.h>
 
 // BKT=64 halves barriers but its 55 KB LDS halves occupancy (2 vs 4 blocks/CU)
 // — the crossover is empirical, so the threshold is runtime-tunable.
@@ -1017,7 +1035,7 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   }
   int M = N * H * W;
   int RSK = R * S * K;
-  int fast_a = (K % 16 == 0);
+  int fast_a = (K % 16 == 0) && stride <= 2;  // FAST path shift-divides
   int use64 = (RSK >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
   int nk = cdiv(RSK, use64 ? 64 : 32);
@@ -1059,7 +1077,7 @@ void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
   ConvGeom g{N, H, W, C, K, R, S, 1, padh, padw, Ho, Wo};
   int M = N * H * W;
   int RSK = R * S * K;
-  int fast_a = (K % 16 == 0);
+  int fast_a = (K % 16 == 0);  // stride fixed at 1 here
   int use64 = (RSK >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
   int nk = cdiv(RSK, use64 ? 64 : 32);
@@ -1117,8 +1135,14 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
                              int K, int R, int S, int stride, int pad, int Ho,
                              int Wo, int nslices, void* stream) {
   (void)mt;  // kept in the ABI for the (cached) im2col table experiments
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
+             magic40(Ho * Wo), magic40(Wo)};
   int M = N * Ho * Wo;
+  if (M >= (1 << 22) || Ho * Wo >= (1 << 14)) {
+    fprintf(stderr, "cilfw_conv2d_bwd_weight: magic-division range exceeded "
+            "(M=%d HoWo=%d)\n", M, Ho * Wo);
+    abort();
+  }
   int CRS = C * R * S;
   int slice_len = cdiv(M, nslices);
   slice_len = cdiv(slice_len, WBK) * WBK;
