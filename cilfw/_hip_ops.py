@@ -36,7 +36,7 @@ _lib.cilfw_error_string.argtypes = [c_i]
 _PROTOS = {
     "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
-    "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 13 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
     "cilfw_im2col_smallc": [c_vp] * 3 + [c_i] * 11 + [c_vp],
     "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
@@ -215,7 +215,10 @@ def _mtable(N, Ho, Wo, stride, device):
     return mt
 
 
-def conv2d_bwd_weight(dy, x, stride, pad, R, S):
+def conv2d_bwd_weight(dy, x, stride, pad, R, S, out=None, accum=False):
+    """dW (R,S,C,K) fp32. With ``out`` the reduce writes straight into the
+    given buffer (the DataParallelEngine's flat-grad slot — no AccumulateGrad
+    add); ``accum=True`` adds instead of overwriting (grad accumulation)."""
     _bf16(dy, "conv2d_bwd_weight.dy")
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
@@ -235,18 +238,28 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
                                      _ptr(dwp), _ptr(ws), c_i(M), c_i(1),
                                      c_i(1), c_i(CRSpad), c_i(K), c_i(1),
                                      c_i(1), c_i(1), c_i(0), c_i(1), c_i(1),
-                                     c_i(ns), _stream())
+                                     c_i(ns), c_i(0), _stream())
         _check("conv2d_bwd_weight_stem")
-        return dwp[:CRS].reshape(R, S, C, K)
+        dw = dwp[:CRS].reshape(R, S, C, K)
+        if out is None:
+            return dw
+        (out.add_(dw) if accum else out.copy_(dw))
+        return out
     if R == 1 and S == 1 and stride > 1:
         xg = torch.empty(N, Ho, Wo, C, dtype=torch.bfloat16, device=x.device)
         _lib.cilfw_stride_gather(_ptr(x), _ptr(xg), c_i(N), c_i(H), c_i(W_),
                                  c_i(C), c_i(stride), c_i(Ho), c_i(Wo),
                                  _stream())
         _check("stride_gather")
-        return conv2d_bwd_weight(dy, xg, 1, 0, R, S)
+        return conv2d_bwd_weight(dy, xg, 1, 0, R, S, out=out, accum=accum)
     mt = _mtable(N, Ho, Wo, stride, dy.device)
-    dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
+    if out is None:
+        dw = torch.empty(R, S, C, K, dtype=torch.float32, device=dy.device)
+        accum = False
+    else:
+        assert out.is_contiguous() and out.dtype == torch.float32 \
+            and out.shape == (R, S, C, K)
+        dw = out
     ns = _lib.cilfw_conv2d_bwd_weight_nslices(N, C, K, R, S, Ho, Wo)
     ws = torch.empty(ns * R * S * C * K, dtype=torch.float32,
                      device=dy.device)
@@ -254,7 +267,7 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S):
                                  _ptr(ws), c_i(N), c_i(H), c_i(W_), c_i(C),
                                  c_i(K), c_i(R), c_i(S), c_i(stride),
                                  c_i(pad), c_i(Ho), c_i(Wo), c_i(ns),
-                                 _stream())
+                                 c_i(1 if accum else 0), _stream())
     _check("conv2d_bwd_weight")
     return dw
 

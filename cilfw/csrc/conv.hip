@@ -62,10 +62,12 @@ void reduce_slabs_bf16_kernel(const float* __restrict__ ws,
   }
 }
 
-// sum ksplit fp32 slabs -> fp32 out (weight grads)
+// sum ksplit fp32 slabs -> fp32 out (weight grads); accum adds into out
+// (grad-accumulation steps deliver straight into the flat grad slot)
 __global__ __launch_bounds__(NTHREADS)
 void reduce_slabs_f32_kernel(const float* __restrict__ ws,
-                             float* __restrict__ out, int ns, long len) {
+                             float* __restrict__ out, int ns, long len,
+                             int accum) {
   long i0 = ((long)blockIdx.x * NTHREADS + threadIdx.x) * 4;
   if (i0 >= len) return;
   if (i0 + 4 <= len) {
@@ -74,12 +76,16 @@ void reduce_slabs_f32_kernel(const float* __restrict__ ws,
       float4 v = *(const float4*)&ws[(long)s * len + i0];
       acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
     }
+    if (accum) {
+      float4 o = *(const float4*)&out[i0];
+      acc.x += o.x; acc.y += o.y; acc.z += o.z; acc.w += o.w;
+    }
     *(float4*)&out[i0] = acc;
   } else {
     for (long i = i0; i < len; ++i) {
       float a = ws[i];
       for (int s = 1; s < ns; ++s) a += ws[(long)s * len + i];
-      out[i] = a;
+      out[i] = accum ? out[i] + a : a;
     }
   }
 }
@@ -1133,7 +1139,7 @@ int cilfw_conv2d_bwd_data_ksplit(int N, int H, int W, int C, int K, int R,
 void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
                              void* dw, void* ws, int N, int H, int W, int C,
                              int K, int R, int S, int stride, int pad, int Ho,
-                             int Wo, int nslices, void* stream) {
+                             int Wo, int nslices, int accum, void* stream) {
   (void)mt;  // kept in the ABI for the (cached) im2col table experiments
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
              magic40(Ho * Wo), magic40(Wo)};
@@ -1162,7 +1168,7 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
   hipLaunchKernelGGL(reduce_slabs_f32_kernel,
                      dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
                      dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
-                     (float*)dw, nslices, len);
+                     (float*)dw, nslices, len, accum);
 }
 
 void cilfw_im2col_smallc(const void* x, const void* mt, void* col,

@@ -80,6 +80,16 @@ class DataParallelEngine:
             self._hooks.append(p.register_post_accumulate_grad_hook(
                 self._make_hook(i)))
 
+        # grad sink: backward kernels that know their param can write the
+        # gradient STRAIGHT into the flat slot (no AccumulateGrad add, and
+        # for conv dW no side-stream join inside backward — the join moves
+        # to finalize()). ops/functional.py consults p._cilfw_sink.
+        self._epoch = 0
+        self._side_dirty = False
+        for i, p in enumerate(self.params):
+            p._cilfw_sink = (self, i)
+            p._cilfw_sink_epoch = -1
+
         # replicate rank-0 initial weights (reference: DDP broadcast at wrap,
         # SURVEY.md §2.3 N6) — one collective for the whole model
         if _dist_active():
@@ -109,23 +119,56 @@ class DataParallelEngine:
         if self._launched[bi] or not _dist_active():
             self._launched[bi] = True
             return
+        self._join_side_stream()  # bucket may hold side-stream dW deliveries
         s, e, _ = self.buckets[bi]
         work = dist.all_reduce(self.flat_grads[s:e], op=dist.ReduceOp.SUM,
                                group=self.group, async_op=True)
         self._works.append(work)
         self._launched[bi] = True
 
+    # ---- grad sink (direct in-kernel delivery into the flat slots) ----
+
+    def sink_acquire(self, i):
+        """(flat-grad view for param i, accumulate?) — accumulate when this
+        param already delivered since the last zero_grad (grad accumulation
+        across no_sync micro-batches)."""
+        p = self.params[i]
+        off = self._offsets[i]
+        view = self.flat_grads[off:off + p.numel()]
+        return view, p._cilfw_sink_epoch == self._epoch
+
+    def sink_delivered(self, i, side_stream=False):
+        """Mark param i's gradient as written (kernel already enqueued)."""
+        p = self.params[i]
+        p._cilfw_sink_epoch = self._epoch
+        if side_stream:
+            self._side_dirty = True
+        if self._hook_enabled:
+            bi = self._param_bucket[i]
+            self._arrived[bi] += 1
+            if self._arrived[bi] == len(self.buckets[bi][2]):
+                self._launch(bi)
+
+    def _join_side_stream(self):
+        if self._side_dirty and self.flat_grads.is_cuda:
+            from ..ops.functional import _wstream
+            torch.cuda.current_stream().wait_stream(_wstream())
+        self._side_dirty = False
+
     def zero_grad(self):
         self.flat_grads.zero_()
         self._arrived = [0] * len(self.buckets)
         self._launched = [False] * len(self.buckets)
         self._works = []
+        self._epoch += 1
 
     def finalize(self):
-        """Call after backward: flush stragglers, wait, average."""
+        """Call after backward: flush stragglers, wait, average; join any
+        side-stream grad deliveries so the optimizer sees complete grads."""
         for bi in range(len(self.buckets)):
             if not self._launched[bi]:
                 self._launch(bi)
+        self._join_side_stream()
         for w in self._works:
             w.wait()
         self._works = []
@@ -153,3 +196,6 @@ class DataParallelEngine:
             p.grad = None
             if hasattr(p, "_cilfw_bf16"):
                 del p._cilfw_bf16
+            if hasattr(p, "_cilfw_sink"):
+                del p._cilfw_sink
+                del p._cilfw_sink_epoch

@@ -58,6 +58,7 @@ class Conv2dNHWC(torch.autograd.Function):
             wc = w.to(x.dtype)
         ctx.save_for_backward(x, wc)
         ctx.w_dtype = w.dtype
+        ctx.w_ref = w  # grad-sink lookup (engine flat-slot delivery)
         if use_hip(x):
             return ext().conv2d_fwd(x, wc, stride, padding)
         # CPU reference: fp32 NCHW conv
@@ -72,13 +73,41 @@ class Conv2dNHWC(torch.autograd.Function):
         stride, padding = ctx.stride, ctx.padding
         dy = dy.contiguous()
         need_dx = ctx.needs_input_grad[0]  # stems skip the whole bwd-data pass
+        # sink modes (A/B-measured, gpurun 20-step bench ×2): off 3.88 ms,
+        # join 3.80 ms, defer 3.84-4.09 ms (erratic stream scheduling) —
+        # direct slot delivery wins, the deferred side-stream join does not
+        import os as _os
+        _mode = _os.environ.get("CILFW_SINK_MODE", "join")
+        sink = (getattr(ctx.w_ref, "_cilfw_sink", None)
+                if _mode != "off" else None)
         if use_hip(dy):
+            R, S = wc.shape[0], wc.shape[1]
             cur = torch.cuda.current_stream()
             ws = _wstream()
             ws.wait_stream(cur)
+            if sink is not None:
+                # write dW straight into the engine's flat-grad slot on the
+                # side stream; no join here — backward keeps going and the
+                # engine joins once in finalize(). No AccumulateGrad add.
+                eng, pi = sink
+                slot, accum = eng.sink_acquire(pi)
+                out = slot.view(R, S, wc.shape[2], wc.shape[3])
+                with torch.cuda.stream(ws):
+                    ext().conv2d_bwd_weight(dy, x, stride, padding, R, S,
+                                            out=out, accum=accum)
+                dy.record_stream(ws)
+                x.record_stream(ws)
+                dx = ext().conv2d_bwd_data(dy, wc, stride, padding,
+                                           x.shape[1], x.shape[2]) \
+                    if need_dx else None
+                if _mode == "join":  # A/B: immediate join as before
+                    cur.wait_stream(ws)
+                    eng.sink_delivered(pi, side_stream=False)
+                else:
+                    eng.sink_delivered(pi, side_stream=True)
+                return dx, None, None, None
             with torch.cuda.stream(ws):  # dW concurrent with dX (fork/join)
-                dw = ext().conv2d_bwd_weight(dy, x, stride, padding,
-                                             wc.shape[0], wc.shape[1])
+                dw = ext().conv2d_bwd_weight(dy, x, stride, padding, R, S)
             dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1],
                                        x.shape[2]) if need_dx else None
             cur.wait_stream(ws)
@@ -95,6 +124,13 @@ class Conv2dNHWC(torch.autograd.Function):
         dwf = torch.nn.grad.conv2d_weight(xf, wf.shape, dyf, stride=stride,
                                           padding=padding)
         dw = dwf.permute(2, 3, 1, 0).contiguous().to(ctx.w_dtype)  # (R,S,C,K)
+        if sink is not None:
+            eng, pi = sink
+            slot, accum = eng.sink_acquire(pi)
+            out = slot.view_as(dw)
+            (out.add_(dw) if accum else out.copy_(dw))
+            eng.sink_delivered(pi)
+            return dx, None, None, None
         return dx, dw, None, None
 
 
