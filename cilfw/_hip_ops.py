@@ -41,7 +41,7 @@ _PROTOS = {
     "cilfw_im2col_smallc": [c_vp] * 3 + [c_i] * 11 + [c_vp],
     "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
     "cilfw_bn_fwd": [c_vp] * 10 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
-    "cilfw_bn_bwd": [c_vp] * 9 + [c_l, c_i, c_i, c_i, c_vp],
+    "cilfw_bn_bwd": [c_vp] * 11 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_downsample_a_fwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
@@ -314,7 +314,11 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
     return y, mean, invstd
 
 
-def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training, want_dres=False):
+def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training, want_dres=False,
+           out_gamma=None, out_beta=None):
+    """With out_gamma/out_beta the reduced per-channel grads are written
+    straight into the given fp32 buffers (flat-grad slot delivery) and the
+    returned dgamma/dbeta are those buffers."""
     _bf16(dy, "bn_bwd.dy")
     C = x.shape[-1]
     M = x.numel() // C
@@ -325,11 +329,14 @@ def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training, want_dres=False):
     gf = gamma.float().contiguous()
     _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(dres),
                       _ptr(gf), _ptr(mean), _ptr(invstd), _ptr(dgb),
+                      _ptr(out_gamma), _ptr(out_beta),
                       c_l(M), c_i(C), c_i(1 if relu else 0),
                       c_i(1 if training else 0), _stream())
     _check("bn_bwd")
     base = gy * 2 * C
-    return dx, dgb[base:base + C], dgb[base + C:base + 2 * C], dres
+    dgamma = out_gamma if out_gamma is not None else dgb[base:base + C]
+    dbeta = out_beta if out_beta is not None else dgb[base + C:base + 2 * C]
+    return dx, dgamma, dbeta, dres
 
 
 # ------------------------------------------------------------------ elementwise

@@ -71,16 +71,23 @@ void bn_sums_kernel(const bf16_t* __restrict__ x, float* __restrict__ part,
 // deterministic slab reduce: out[i] = sum_s ws[s][i]. One WAVE per output
 // element — 64 lanes stride the slab axis in parallel, then a fixed-order
 // shfl tree (ns can be ~512: a thread-serial loop would be pure latency).
+// out2/split: elements i >= split land in out2[i-split] — lets the backward
+// write dgamma and dbeta straight into two separate flat-grad slots
+// (grad-sink delivery; cilfw/distributed/ddp.py)
 __global__ __launch_bounds__(NT)
 void bn_reduce_slabs_kernel(const float* __restrict__ ws,
-                            float* __restrict__ out, int ns, long len) {
+                            float* __restrict__ out, int ns, long len,
+                            float* __restrict__ out2, long split) {
   const int lane = threadIdx.x & 63;
   long i = (long)blockIdx.x * (NT / WAVE) + (threadIdx.x >> 6);
   if (i >= len) return;
   float a = 0.f;
   for (int s = lane; s < ns; s += WAVE) a += ws[(long)s * len + i];
   for (int o = 32; o > 0; o >>= 1) a += __shfl_xor(a, o);
-  if (lane == 0) out[i] = a;
+  if (lane == 0) {
+    if (out2 != nullptr && i >= split) out2[i - split] = a;
+    else out[i] = a;
+  }
 }
 
 // fused slab-reduce + finalize for TRAINING: one wave per channel reduces the
@@ -561,7 +568,8 @@ void cilfw_bn_fwd(const void* x, void* y, const void* res,
                        (const bf16_t*)x, part, M, C, rows_per_blk);
     hipLaunchKernelGGL(bn_reduce_slabs_kernel,
                        dim3(cdiv(2 * C, NT / WAVE)), dim3(NT), 0, st, part,
-                       sum, (int)grid.y, (long)2 * C);
+                       sum, (int)grid.y, (long)2 * C, (float*)nullptr,
+                       (long)2 * C);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv(C, 256)), dim3(256), 0,
                        st, sum, sumsq, (float*)mean, (float*)invstd,
                        (float*)running_mean, (float*)running_var, M, C,
@@ -597,22 +605,24 @@ void cilfw_bn_apply_only(const void* x, void* y, const void* res,
 
 void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                   void* dres, const void* gamma, const void* mean,
-                  const void* invstd, void* dgb, long M, int C, int relu,
-                  int training, void* stream) {
-  // dgb: [gy][2][C] partials followed by the reduced [dgamma | dbeta]
+                  const void* invstd, void* dgb, void* dg_out, void* db_out,
+                  long M, int C, int relu, int training, void* stream) {
+  // dgb: [gy][2][C] partials followed by the reduced [dgamma | dbeta];
+  // dg_out/db_out (optional) divert the reduced grads into flat-grad slots
   hipStream_t st = (hipStream_t)stream;
   int rows_per_blk = 256;
   dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
   float* part = (float*)dgb;
-  float* dgamma = part + (long)grid.y * 2 * C;
-  float* dbeta = dgamma + C;
+  float* dgamma = dg_out ? (float*)dg_out : part + (long)grid.y * 2 * C;
+  float* dbeta = db_out ? (float*)db_out
+                        : part + (long)grid.y * 2 * C + C;
   hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
                      (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
                      (const float*)mean, (const float*)invstd, part,
                      M, C, rows_per_blk, relu);
   hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT / WAVE)),
                      dim3(NT), 0, st, part, dgamma, (int)grid.y,
-                     (long)2 * C);
+                     (long)2 * C, dbeta, (long)C);
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),

@@ -45,20 +45,36 @@ def _to_nhwc(x):
 # --------------------------------------------------------------------------- conv2d
 
 
+def _sink_on():
+    import os
+    return os.environ.get("CILFW_SINK_MODE", "join") != "off"
+
+
 class Conv2dNHWC(torch.autograd.Function):
-    """2D convolution, NHWC activations, (R,S,C,K) weight, symmetric padding."""
+    """2D convolution, NHWC activations, (R,S,C,K) weight, symmetric padding.
+
+    ``w_param`` (non-tensor arg) is set when the weight is engine-registered
+    for grad-sink delivery: ``w`` is then the DETACHED weight, so autograd
+    creates no AccumulateGrad edge for it (returning None would otherwise
+    still materialize a zero grad, run an accumulation kernel and fire the
+    post-accumulate hook — double-counting bucket arrivals), and backward
+    writes dW straight into the engine's flat slot."""
 
     @staticmethod
-    def forward(ctx, x, w, stride, padding):
+    def forward(ctx, x, w, stride, padding, w_param=None):
         ctx.stride, ctx.padding = stride, padding
+        if w_param is not None:
+            w_param = w_param[0]  # holder list: keeps the param OUT of the graph
+        wp = w_param if w_param is not None else w
         # engine-maintained bf16 mirror (updated inside the fused SGD kernel)
         # avoids a per-step cast of the fp32 master
-        wc = getattr(w, "_cilfw_bf16", None)
+        wc = getattr(wp, "_cilfw_bf16", None)
         if wc is None or wc.dtype != x.dtype:
             wc = w.to(x.dtype)
         ctx.save_for_backward(x, wc)
         ctx.w_dtype = w.dtype
-        ctx.w_ref = w  # grad-sink lookup (engine flat-slot delivery)
+        ctx.w_ref = wp  # grad-sink lookup (engine flat-slot delivery)
+        ctx.sinked = w_param is not None
         if use_hip(x):
             return ext().conv2d_fwd(x, wc, stride, padding)
         # CPU reference: fp32 NCHW conv
@@ -73,13 +89,7 @@ class Conv2dNHWC(torch.autograd.Function):
         stride, padding = ctx.stride, ctx.padding
         dy = dy.contiguous()
         need_dx = ctx.needs_input_grad[0]  # stems skip the whole bwd-data pass
-        # sink modes (A/B-measured, gpurun 20-step bench ×2): off 3.88 ms,
-        # join 3.80 ms, defer 3.84-4.09 ms (erratic stream scheduling) —
-        # direct slot delivery wins, the deferred side-stream join does not
-        import os as _os
-        _mode = _os.environ.get("CILFW_SINK_MODE", "join")
-        sink = (getattr(ctx.w_ref, "_cilfw_sink", None)
-                if _mode != "off" else None)
+        sink = getattr(ctx.w_ref, "_cilfw_sink", None) if ctx.sinked else None
         if use_hip(dy):
             R, S = wc.shape[0], wc.shape[1]
             cur = torch.cuda.current_stream()
@@ -87,32 +97,26 @@ class Conv2dNHWC(torch.autograd.Function):
             ws.wait_stream(cur)
             if sink is not None:
                 # write dW straight into the engine's flat-grad slot on the
-                # side stream; no join here — backward keeps going and the
-                # engine joins once in finalize(). No AccumulateGrad add.
+                # side stream (no fresh tensor, no AccumulateGrad add)
                 eng, pi = sink
                 slot, accum = eng.sink_acquire(pi)
                 out = slot.view(R, S, wc.shape[2], wc.shape[3])
                 with torch.cuda.stream(ws):
                     ext().conv2d_bwd_weight(dy, x, stride, padding, R, S,
                                             out=out, accum=accum)
-                dy.record_stream(ws)
-                x.record_stream(ws)
                 dx = ext().conv2d_bwd_data(dy, wc, stride, padding,
                                            x.shape[1], x.shape[2]) \
                     if need_dx else None
-                if _mode == "join":  # A/B: immediate join as before
-                    cur.wait_stream(ws)
-                    eng.sink_delivered(pi, side_stream=False)
-                else:
-                    eng.sink_delivered(pi, side_stream=True)
-                return dx, None, None, None
+                cur.wait_stream(ws)
+                eng.sink_delivered(pi)
+                return dx, None, None, None, None
             with torch.cuda.stream(ws):  # dW concurrent with dX (fork/join)
                 dw = ext().conv2d_bwd_weight(dy, x, stride, padding, R, S)
             dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1],
                                        x.shape[2]) if need_dx else None
             cur.wait_stream(ws)
             dw.record_stream(cur)
-            return dx, dw.to(ctx.w_dtype), None, None
+            return dx, dw.to(ctx.w_dtype), None, None, None
         xf = _to_nchw(x).float()
         wf = wc.permute(3, 2, 0, 1).contiguous().float()
         dyf = _to_nchw(dy).float()
@@ -130,15 +134,59 @@ class Conv2dNHWC(torch.autograd.Function):
             out = slot.view_as(dw)
             (out.add_(dw) if accum else out.copy_(dw))
             eng.sink_delivered(pi)
-            return dx, None, None, None
-        return dx, dw, None, None
+            return dx, None, None, None, None
+        return dx, dw, None, None, None
 
 
 def conv2d(x, w, stride=1, padding=1):
+    # engine-registered weights take the grad-sink path: DETACHED weight in
+    # the graph + the param as a non-tensor ref (x must carry requires_grad
+    # or the graph would die at input-adjacent convs, e.g. the stem)
+    if (_sink_on() and x.requires_grad
+            and getattr(w, "_cilfw_sink", None) is not None):
+        return Conv2dNHWC.apply(x, w.detach(), stride, padding, [w])
     return Conv2dNHWC.apply(x, w, stride, padding)
 
 
 # ----------------------------------------------------------------- batchnorm (+ReLU)
+
+
+def _gb_sinkable(x, gamma, beta):
+    """True when both BN params are engine-registered for grad-sink delivery
+    (and x carries the graph, so detaching them cannot kill it)."""
+    if not (_sink_on() and x.requires_grad):
+        return False
+    sg = getattr(gamma, "_cilfw_sink", None)
+    sb = getattr(beta, "_cilfw_sink", None)
+    return sg is not None and sb is not None and sg[0] is sb[0]
+
+
+def _gb_deliver_hip(eng, gi, bi, bn_bwd_args, want_dres):
+    """BN backward with dgamma/dbeta written into the flat slots. During
+    grad accumulation the reduced grads go to scratch first (the dx formula
+    needs THIS batch's dgamma/dbeta, not the accumulated slot) and are added."""
+    gs, gacc = eng.sink_acquire(gi)
+    bs, bacc = eng.sink_acquire(bi)
+    if gacc or bacc:
+        dx, dgamma, dbeta, dres = ext().bn_bwd(*bn_bwd_args,
+                                               want_dres=want_dres)
+        gs.add_(dgamma)
+        bs.add_(dbeta)
+    else:
+        dx, _, _, dres = ext().bn_bwd(*bn_bwd_args, want_dres=want_dres,
+                                      out_gamma=gs, out_beta=bs)
+    eng.sink_delivered(gi)
+    eng.sink_delivered(bi)
+    return dx, dres
+
+
+def _gb_deliver_cpu(eng, gi, bi, dgamma, dbeta):
+    gs, gacc = eng.sink_acquire(gi)
+    bs, bacc = eng.sink_acquire(bi)
+    (gs.add_(dgamma) if gacc else gs.copy_(dgamma))
+    (bs.add_(dbeta) if bacc else bs.copy_(dbeta))
+    eng.sink_delivered(gi)
+    eng.sink_delivered(bi)
 
 
 class BatchNormAct(torch.autograd.Function):
@@ -151,9 +199,17 @@ class BatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
-                training, relu):
+                training, relu, g_param=None, b_param=None):
         ctx.relu = relu
         ctx.eps = eps
+        # grad-sink: g_param/b_param set => gamma/beta are DETACHED (see
+        # Conv2dNHWC docstring for why returning None is not enough); the
+        # params arrive in holder lists so apply() does not graph-track them
+        if g_param is not None:
+            g_param, b_param = g_param[0], b_param[0]
+        ctx.g_ref = g_param if g_param is not None else gamma
+        ctx.b_ref = b_param if b_param is not None else beta
+        ctx.sinked = g_param is not None
         if use_hip(x):
             y, save_mean, save_invstd = ext().bn_fwd(
                 x, gamma, beta, running_mean, running_var,
@@ -187,10 +243,20 @@ class BatchNormAct(torch.autograd.Function):
         x, gamma, mean, invstd, y = ctx.saved_tensors
         relu = ctx.relu
         dy = dy.contiguous()
+        sink = (getattr(ctx.g_ref, "_cilfw_sink", None)
+                if ctx.sinked else None)
         if use_hip(dy):
+            if sink is not None:  # reduce writes straight into the slots
+                eng, gi = sink
+                bi = ctx.b_ref._cilfw_sink[1]
+                dx, _ = _gb_deliver_hip(
+                    eng, gi, bi,
+                    (dy, x, gamma, mean, invstd, y, relu, ctx.training),
+                    want_dres=False)
+                return (dx, None, None) + (None,) * 8
             dx, dgamma, dbeta, _ = ext().bn_bwd(dy, x, gamma, mean, invstd,
                                                 y, relu, ctx.training)
-            return (dx, dgamma, dbeta) + (None,) * 6
+            return (dx, dgamma, dbeta) + (None,) * 8
         C = x.shape[-1]
         dyf = dy.float().reshape(-1, C)
         if relu:
@@ -206,11 +272,20 @@ class BatchNormAct(torch.autograd.Function):
         else:
             dxf = dyf * gamma.float() * invstd
         dx = dxf.reshape_as(x).to(x.dtype)
-        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype)) + (None,) * 6
+        if sink is not None:
+            eng, gi = sink
+            bi = ctx.b_ref._cilfw_sink[1]
+            _gb_deliver_cpu(eng, gi, bi, dgamma, dbeta)
+            return (dx, None, None) + (None,) * 8
+        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype)) + (None,) * 8
 
 
 def batchnorm_act(x, gamma, beta, running_mean, running_var, momentum=0.1, eps=1e-5,
                   training=True, relu=False):
+    if _gb_sinkable(x, gamma, beta):
+        return BatchNormAct.apply(x, gamma.detach(), beta.detach(),
+                                  running_mean, running_var, momentum, eps,
+                                  training, relu, [gamma], [beta])
     return BatchNormAct.apply(x, gamma, beta, running_mean, running_var, momentum,
                               eps, training, relu)
 
@@ -222,8 +297,13 @@ class BatchNormAddReLU(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
-                momentum, eps, training):
+                momentum, eps, training, g_param=None, b_param=None):
         ctx.eps = eps
+        if g_param is not None:  # holder lists (see BatchNormAct)
+            g_param, b_param = g_param[0], b_param[0]
+        ctx.g_ref = g_param if g_param is not None else gamma
+        ctx.b_ref = b_param if b_param is not None else beta
+        ctx.sinked = g_param is not None
         if use_hip(x):
             y, save_mean, save_invstd = ext().bn_fwd(
                 x, gamma, beta, running_mean, running_var, momentum, eps,
@@ -255,11 +335,21 @@ class BatchNormAddReLU(torch.autograd.Function):
     def backward(ctx, dy):
         x, gamma, mean, invstd, y = ctx.saved_tensors
         dy = dy.contiguous()
+        sink = (getattr(ctx.g_ref, "_cilfw_sink", None)
+                if ctx.sinked else None)
         if use_hip(dy):
+            if sink is not None:  # reduce writes straight into the slots
+                eng, gi = sink
+                bi = ctx.b_ref._cilfw_sink[1]
+                dx, dres = _gb_deliver_hip(
+                    eng, gi, bi,
+                    (dy, x, gamma, mean, invstd, y, True, ctx.training),
+                    want_dres=True)
+                return (dx, dres, None, None) + (None,) * 7
             dx, dgamma, dbeta, dres = ext().bn_bwd(
                 dy, x, gamma, mean, invstd, y, True, ctx.training,
                 want_dres=True)
-            return (dx, dres, dgamma, dbeta) + (None,) * 5
+            return (dx, dres, dgamma, dbeta) + (None,) * 7
         C = x.shape[-1]
         dyf = dy.float().reshape(-1, C) * (y.float().reshape(-1, C) > 0)
         xf = x.float().reshape(-1, C)
@@ -273,12 +363,23 @@ class BatchNormAddReLU(torch.autograd.Function):
         else:
             dxf = dyf * gamma.float() * invstd
         dres = dyf.reshape_as(x).to(dy.dtype)
+        if sink is not None:
+            eng, gi = sink
+            bi = ctx.b_ref._cilfw_sink[1]
+            _gb_deliver_cpu(eng, gi, bi, dgamma, dbeta)
+            return (dxf.reshape_as(x).to(x.dtype), dres, None,
+                    None) + (None,) * 7
         return (dxf.reshape_as(x).to(x.dtype), dres, dgamma.to(gamma.dtype),
-                dbeta.to(gamma.dtype)) + (None,) * 5
+                dbeta.to(gamma.dtype)) + (None,) * 7
 
 
 def batchnorm_add_relu(x, residual, gamma, beta, running_mean, running_var,
                        momentum=0.1, eps=1e-5, training=True):
+    if _gb_sinkable(x, gamma, beta):
+        return BatchNormAddReLU.apply(x, residual, gamma.detach(),
+                                      beta.detach(), running_mean,
+                                      running_var, momentum, eps, training,
+                                      [gamma], [beta])
     return BatchNormAddReLU.apply(x, residual, gamma, beta, running_mean,
                                   running_var, momentum, eps, training)
 
