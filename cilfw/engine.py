@@ -271,7 +271,14 @@ def extract_task_features(model, dataset, device, args):
                                shuffle=False, augment=False, drop_last=False,
                                dtype=dtype)
     else:
-        loader = DataLoader(dataset, batch_size=args.batch_size,
+        # un-augmented features on the CPU path too (the device path always
+        # extracts with augment=False; advisor round-1 flagged the mismatch)
+        from .data.scenario import TaskSet
+        from .data.transforms import EvalTransform
+        ds_eval = TaskSet(dataset.x, dataset.y, dataset.t,
+                          EvalTransform(args, getattr(args, "_stats_key",
+                                                      "synthetic")))
+        loader = DataLoader(ds_eval, batch_size=args.batch_size,
                             shuffle=False, num_workers=args.workers,
                             drop_last=False)
     feats = []
@@ -350,11 +357,26 @@ def run(args):
         if use_gpu_data:
             mean, std = DATASET_STATS[getattr(args, "_stats_key", "synthetic")]
             train_sampler = None
+            # full device-side recipe: RandAugment/jitter pre-normalize +
+            # random erasing post-normalize (cilfw/data/device_augment.py);
+            # honors the same --aa/--color_jitter/--reprob flags as the host
+            # pipeline (the round-1 device path silently dropped them)
+            aug = None
+            if not args.no_aug and (getattr(args, "aa", "")
+                                    or getattr(args, "color_jitter", 0)
+                                    or getattr(args, "reprob", 0)):
+                from .data.device_augment import DeviceAugment
+                aug = DeviceAugment(
+                    aa_policy=getattr(args, "aa", ""),
+                    color_jitter=getattr(args, "color_jitter", 0.0),
+                    reprob=getattr(args, "reprob", 0.0),
+                    remode=getattr(args, "remode", "pixel"),
+                    recount=getattr(args, "recount", 1))
             train_loader = GpuTaskLoader(
                 dataset_train, args.batch_size, device, mean, std,
                 world=world, rank=rank, shuffle=True, seed=args.seed,
                 augment=not args.no_aug, drop_last=True,
-                dtype=compute_dtype(args), extra=extra)
+                dtype=compute_dtype(args), extra=extra, aug_pipeline=aug)
             val_loader = GpuTaskLoader(
                 dataset_val, args.batch_size, device, mean, std,
                 world=world, rank=rank, shuffle=False, augment=False,
