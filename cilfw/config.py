@@ -114,6 +114,11 @@ def get_args_parser():
     parser.add_argument("--no_step_graph", action="store_true", default=False,
                         help="disable hipGraph capture of the training step "
                              "(graphs are on by default with --gpu_data)")
+    parser.add_argument("--dist_timeout", default=300.0, type=float,
+                        help="collective timeout in seconds — a dead rank "
+                             "raises instead of hanging (reference behavior "
+                             "is an indefinite hang); also drives the "
+                             "heartbeat watchdog")
     parser.add_argument("--compat_step_barrier", action="store_true", default=False,
                         help="reproduce the reference's per-training-step "
                              "dist.barrier (template.py:272) — a perf bug kept "

@@ -52,10 +52,16 @@ def init_distributed_mode(args):
         backend = "gloo"
         if getattr(args, "device", None) is None:
             args.device = "cpu"
+    # finite collective timeout — a dead peer surfaces as an error instead
+    # of hanging forever (the reference hangs, template.py:272). For RCCL
+    # the NCCL watchdog needs async error handling to turn a stuck
+    # collective into an exception.
+    timeout_s = float(getattr(args, "dist_timeout", 300.0) or 300.0)
+    os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
     dist.init_process_group(backend=backend,
                             init_method=getattr(args, "dist_url", "env://"),
                             world_size=args.world_size, rank=args.rank,
-                            timeout=datetime.timedelta(minutes=30))
+                            timeout=datetime.timedelta(seconds=timeout_s))
     dist.barrier()
     setup_for_distributed(args.rank == 0)
 

@@ -214,7 +214,10 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         first_of_epoch = True
         metric_every = max(int(getattr(args, "metric_every", 1)), 1)
         step_i = 0
+        wd = getattr(args, "_watchdog", None)
         for inputs, targets, _tids in train_loader:
+            if wd is not None:
+                wd.beat()
             inputs, targets = _to_device(inputs, targets, device, dtype)
             if can_graph and graphed is None and not first_of_epoch:
                 # capture on this batch; the capture run IS its training step.
@@ -293,6 +296,36 @@ def run(args):
     init_distributed_mode(args)
     init_seed(args)
     device = args.device
+
+    # failure detection (reference has none — a dead rank hangs the job):
+    # collectives time out via the process-group timeout; the heartbeat
+    # watchdog covers non-collective hangs. Watchdog fires a bit after the
+    # collective timeout so the cleaner error path wins when both apply.
+    from .distributed.watchdog import Watchdog, describe_failure
+    watchdog = None
+    if get_world_size() > 1:
+        watchdog = Watchdog(
+            float(getattr(args, "dist_timeout", 300.0) or 300.0) * 1.5,
+            rank=get_rank(), checkpoint_dir=args.output_dir).start()
+    args._watchdog = watchdog
+    try:
+        return _run_tasks(args, device, watchdog)
+    except RuntimeError as e:
+        if watchdog is not None:
+            watchdog.stop()
+        if get_world_size() > 1:
+            import sys
+            print(describe_failure(
+                e, watchdog.last_checkpoint if watchdog else None),
+                file=sys.stderr, flush=True)
+            raise SystemExit(3)
+        raise
+    finally:
+        if watchdog is not None:
+            watchdog.stop()
+
+
+def _run_tasks(args, device, watchdog):
 
     scenario_train, nb_classes = build_dataset(is_train=True, args=args)
     scenario_val, _ = build_dataset(is_train=False, args=args)
@@ -432,8 +465,10 @@ def run(args):
         engine.detach()
         args.known_classes += args.increment_per_task  # before snapshot: the
         # checkpoint records the post-task state so resume starts task t+1
-        save_task_checkpoint(args.output_dir, task_id, model, memory, acc1s,
-                             args, optimizer, scheduler)
+        ckpt = save_task_checkpoint(args.output_dir, task_id, model, memory,
+                                    acc1s, args, optimizer, scheduler)
+        if watchdog is not None and ckpt:
+            watchdog.note_checkpoint(ckpt)
 
     avg_inc_acc = sum(acc1s) / len(acc1s) if acc1s else 0.0
     print(f"average incremental accuracy = {avg_inc_acc:.5f}")
