@@ -63,7 +63,7 @@ _PROTOS = {
     "cilfw_kd_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
     "cilfw_wa_loss_fwd": [c_vp] * 8 + [c_i] * 3 + [c_f] * 3 + [c_vp],
     "cilfw_wa_loss_bwd": [c_vp] * 6 + [c_i] * 3 + [c_f] * 3 + [c_vp],
-    "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_f, c_f, c_f, c_vp],
+    "cilfw_sgd_step": [c_vp] * 4 + [c_l, c_vp, c_f, c_f, c_vp],
     "cilfw_topk_correct": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_herding_select": [c_vp] * 3 + [c_i] * 3 + [c_vp],
     "cilfw_herding_select_batch": [c_vp] * 6 + [c_i] * 3 + [c_vp],
@@ -538,9 +538,14 @@ def wa_loss_bwd(probs, ps, pt, targets, dtotal, smooth, T, lam, Ck):
 
 # ------------------------------------------------------------- optimizer / misc
 
-def sgd_step(p, g, m, lr, momentum, wd, p_bf16=None):
+def sgd_step(p, g, m, lr_dev, momentum, wd, p_bf16=None):
+    """lr_dev: 1-elem fp32 DEVICE tensor (graph-replayable lr input);
+    a plain float is boxed into one for convenience."""
+    if not torch.is_tensor(lr_dev):
+        lr_dev = torch.full((1,), float(lr_dev), dtype=torch.float32,
+                            device=p.device)
     _lib.cilfw_sgd_step(_ptr(p), _ptr(g), _ptr(m), _ptr(p_bf16),
-                        c_l(p.numel()), c_f(lr), c_f(momentum), c_f(wd),
+                        c_l(p.numel()), _ptr(lr_dev), c_f(momentum), c_f(wd),
                         _stream())
     _check("sgd_step")
 

@@ -301,10 +301,14 @@ void wa_loss_bwd_kernel(const float* __restrict__ probs,
 // ------------------------------------------------------------------- fused SGD
 // g = grad + wd*p; m = mu*m + g; p -= lr*m   — one kernel over the flat buffers.
 
+// lr arrives through DEVICE memory (lr_dev) so a hipGraph-captured step can
+// be re-used across epochs: the cosine scheduler writes the scalar once per
+// epoch instead of forcing a re-capture (the lr would otherwise be baked in)
 __global__ __launch_bounds__(NT)
 void sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
                 float* __restrict__ m, bf16_t* __restrict__ pb, long n,
-                float lr, float mu, float wd) {
+                const float* __restrict__ lr_dev, float mu, float wd) {
+  const float lr = lr_dev[0];
   long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 4;
   if (i0 >= n) return;
   if (i0 + 4 <= n) {
@@ -565,11 +569,11 @@ void cilfw_kd_bwd(const void* ps, const void* pt, const void* dloss, void* ds,
 }
 
 void cilfw_sgd_step(void* p, const void* g, void* m, void* pb, long n,
-                    float lr, float mu, float wd, void* stream) {
+                    const void* lr_dev, float mu, float wd, void* stream) {
   long blocks = cdiv((long)n, (long)NT * 4);
   hipLaunchKernelGGL(sgd_kernel, dim3((int)blocks), dim3(NT), 0,
                      (hipStream_t)stream, (float*)p, (const float*)g,
-                     (float*)m, (bf16_t*)pb, n, lr, mu, wd);
+                     (float*)m, (bf16_t*)pb, n, (const float*)lr_dev, mu, wd);
 }
 
 void cilfw_topk_correct(const void* logits, const void* targets, void* counts,

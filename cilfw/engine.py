@@ -199,6 +199,9 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
                  and (get_world_size() == 1
                       or _os.environ.get("CILFW_GRAPH_MULTI") == "1"))
     model.train()
+    graphed = None  # ONE capture per task: the lr is a device-tensor input
+    # to the captured SGD launch (FlatSGD._lr_dev), so the cosine schedule
+    # updates it by a 4-byte fill instead of forcing per-epoch re-capture
     for epoch in range(args.num_epochs):
         if train_sampler is not None:
             train_sampler.set_epoch(epoch)
@@ -210,8 +213,7 @@ def train_one_task(model, teacher, engine, optimizer, scheduler, train_loader,
         metric_logger.meters["lr"] = SmoothedValue(fmt="{value:.6f}")
         t0 = time.time()
         nimg = 0
-        graphed = None  # per-epoch: the cosine LR is baked into the capture
-        first_of_epoch = True
+        first_of_epoch = epoch == 0
         metric_every = max(int(getattr(args, "metric_every", 1)), 1)
         step_i = 0
         wd = getattr(args, "_watchdog", None)

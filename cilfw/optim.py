@@ -18,11 +18,27 @@ from .ops._backend import use_hip, ext
 class FlatSGD:
     def __init__(self, engine, lr, momentum=0.9, weight_decay=5e-4):
         self.engine = engine
-        self.lr = lr
+        # lr lives in a 1-elem DEVICE tensor on GPU: the hipGraph-captured
+        # step reads it through memory, so the cosine scheduler only writes
+        # the scalar per epoch instead of forcing a per-epoch re-capture
+        self._lr_dev = (torch.full((1,), float(lr), dtype=torch.float32,
+                                   device=engine.flat_params.device)
+                        if engine.flat_params.is_cuda else None)
+        self._lr = float(lr)
         self.base_lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
         self.momentum_buf = torch.zeros_like(engine.flat_grads)
+
+    @property
+    def lr(self):
+        return self._lr
+
+    @lr.setter
+    def lr(self, v):
+        self._lr = float(v)
+        if self._lr_dev is not None:
+            self._lr_dev.fill_(float(v))
 
     @torch.no_grad()
     def step(self):
@@ -30,8 +46,8 @@ class FlatSGD:
                    self.momentum_buf)
         mirror = getattr(self.engine, "flat_bf16", None)
         if use_hip(p):
-            ext().sgd_step(p, g, m, self.lr, self.momentum, self.weight_decay,
-                           mirror)
+            ext().sgd_step(p, g, m, self._lr_dev, self.momentum,
+                           self.weight_decay, mirror)
             return
         # CPU reference (and CILFW_FORCE_TORCH): same math, fused ATen ops
         if self.weight_decay != 0:
