@@ -63,8 +63,9 @@ def test_conv_fwd_bwd(cin, cout, k, stride, hw, batch):
     dy = torch.randn(yc.shape).to(torch.bfloat16)
     yg.backward(dy.cuda())
     yc.backward(dy)
-    if cin >= 16:  # dx for the generic stem path is unused in practice
-        _cmp(xg.grad, xc.grad, what="conv_dx")
+    # incl. tiny-C stems: the generic bwd-data path is reachable (an input
+    # that itself requires grad), so it is tested, not skipped
+    _cmp(xg.grad, xc.grad, rtol=0.05, atol=0.05, what="conv_dx")
     _cmp(wg32.grad, wc32.grad, rtol=0.03, atol=0.03, what="conv_dw")
 
 
