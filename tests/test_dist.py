@@ -121,3 +121,101 @@ def test_engine_broadcasts_rank0_weights():
     out = mgr.dict()
     mp.spawn(_run_broadcast_worker, args=(2, 29615, out), nprocs=2, join=True)
     assert torch.equal(out[0], out[1])
+
+
+def _run_bucket_fuzz_worker(rank, world, port, out, bucket_mb):
+    _init(rank, world, port)
+    torch.manual_seed(42)
+    model = CilModel("resnet20", 32)
+    model.prev_model_adaption(4)
+    engine = DataParallelEngine(model, bucket_mb=bucket_mb)
+    opt = FlatSGD(engine, lr=0.1, momentum=0.9, weight_decay=0.0)
+    g = torch.Generator().manual_seed(7)
+    x = torch.randn(8, 16, 16, 3, generator=g)
+    y = torch.randint(0, 4, (8,), generator=g)
+    opt.zero_grad()
+    logits, _ = model(x[rank * 4:(rank + 1) * 4])
+    loss = ops.cross_entropy(logits.float(), y[rank * 4:(rank + 1) * 4])
+    loss.backward()
+    engine.finalize()
+    out[(rank, bucket_mb)] = engine.flat_grads.clone()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_bucket_boundary_fuzz():
+    """Averaged gradients must be identical for ANY bucket size — one huge
+    bucket, many tiny ones, and sizes that cut mid-layer."""
+    mgr = mp.Manager()
+    out = mgr.dict()
+    sizes = [100.0, 0.5, 0.037, 0.011, 0.003]
+    for i, mb in enumerate(sizes):
+        mp.spawn(_run_bucket_fuzz_worker, args=(2, 29661 + i, out, mb),
+                 nprocs=2, join=True)
+    ref = out[(0, sizes[0])]
+    for mb in sizes:
+        assert torch.allclose(out[(0, mb)], out[(1, mb)], atol=1e-6), mb
+        assert torch.allclose(out[(0, mb)], ref, atol=1e-6), mb
+
+
+def _run_nosync_worker(rank, world, port, out):
+    _init(rank, world, port)
+    torch.manual_seed(42)
+    model = CilModel("resnet20", 32)
+    model.prev_model_adaption(4)
+    engine = DataParallelEngine(model, bucket_mb=0.05)
+    opt = FlatSGD(engine, lr=0.1, momentum=0.9, weight_decay=0.0)
+    g = torch.Generator().manual_seed(9)
+    xs = [torch.randn(8, 16, 16, 3, generator=g) for _ in range(2)]
+    ys = [torch.randint(0, 4, (8,), generator=g) for _ in range(2)]
+    # 2 accumulation micro-batches under no_sync, then a synced one
+    opt.zero_grad()
+    with engine.no_sync():
+        for x, y in zip(xs, ys):
+            logits, _ = model(x[rank * 4:(rank + 1) * 4])
+            ops.cross_entropy(logits.float(),
+                              y[rank * 4:(rank + 1) * 4]).backward()
+    accum = engine.flat_grads.clone()
+    engine.finalize()  # flushes + averages the accumulated grads
+    out[rank] = {"accum": accum, "avg": engine.flat_grads.clone()}
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_no_sync_accumulates_then_syncs():
+    """Grad accumulation: per-rank sums under no_sync (sink accumulate mode),
+    then one finalize averages across ranks."""
+    mgr = mp.Manager()
+    out = mgr.dict()
+    mp.spawn(_run_nosync_worker, args=(2, 29671, out), nprocs=2, join=True)
+    # pre-finalize grads are rank-local sums => DIFFERENT across ranks
+    assert not torch.allclose(out[0]["accum"], out[1]["accum"], atol=1e-6)
+    # post-finalize: identical averages
+    assert torch.allclose(out[0]["avg"], out[1]["avg"], atol=1e-6)
+    expect = (out[0]["accum"] + out[1]["accum"]) / 2
+    assert torch.allclose(out[0]["avg"], expect, atol=1e-5)
+
+
+def _run_uneven_eval_worker(rank, world, port, out):
+    _init(rank, world, port)
+    from cilfw.data.sampler import DistributedSampler
+
+    class _DS:
+        def __len__(self):
+            return 7  # NOT divisible by world=2 -> pad-by-repetition
+
+    s = DistributedSampler(_DS(), world, rank, shuffle=False)
+    out[rank] = list(iter(s))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_uneven_shard_padding_across_ranks():
+    mgr = mp.Manager()
+    out = mgr.dict()
+    mp.spawn(_run_uneven_eval_worker, args=(2, 29681, out), nprocs=2,
+             join=True)
+    allidx = sorted(out[0] + out[1])
+    assert len(out[0]) == len(out[1]) == 4  # equal shards
+    assert set(allidx) == set(range(7))     # every sample covered
+    assert allidx == [0, 0, 1, 2, 3, 4, 5, 6]  # torch pad-from-front
