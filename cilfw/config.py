@@ -62,7 +62,7 @@ def get_args_parser():
     # dataset (reference template.py:45-46)
     parser.add_argument("--data_set", default="cifar100", type=str,
                         choices=["cifar100", "imagenet100", "imagenet1000", "cub200",
-                                 "synthetic"])
+                                 "synthetic", "synthetic_hard"])
     parser.add_argument("--data_path", default="./data/cifar100", type=str)
 
     # distillation (reference template.py:47-48)
@@ -100,6 +100,11 @@ def get_args_parser():
                         help="HBM-resident task data + on-device batch "
                              "assembly (crop/flip/normalize) — bypasses the "
                              "Python DataLoader for array-backed datasets")
+    parser.add_argument("--no_wa", action="store_true", default=False,
+                        help="ablation: skip the weight-align step")
+    parser.add_argument("--no_replay", action="store_true", default=False,
+                        help="ablation: no rehearsal exemplars (forgetting "
+                             "baseline)")
     parser.add_argument("--no_device_replay", action="store_true", default=False,
                         help="with --gpu_data, source replay via host "
                              "add_samples instead of the HBM-resident "

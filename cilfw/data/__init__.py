@@ -1,7 +1,8 @@
 import numpy as np
 
 from .scenario import ClassIncremental, TaskSet
-from .datasets import build_source, make_synthetic, DATASET_STATS
+from .datasets import (build_source, make_synthetic,
+                       make_synthetic_hard, DATASET_STATS)
 from .transforms import TrainTransform, EvalTransform
 from .sampler import DistributedSampler
 
@@ -44,5 +45,6 @@ def build_dataset(is_train, args, transform="auto"):
 
 
 __all__ = ["ClassIncremental", "TaskSet", "build_source", "make_synthetic",
+           "make_synthetic_hard",
            "build_dataset", "TrainTransform", "EvalTransform",
            "DistributedSampler", "DATASET_STATS", "CIFAR100_CLASS_ORDER"]

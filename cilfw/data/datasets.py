@@ -58,6 +58,52 @@ def make_synthetic(num_classes=100, per_class=50, size=32, seed=0, channels=3):
     return x[perm], y[perm]
 
 
+def make_synthetic_hard(num_classes=100, per_class=50, size=32, seed=0,
+                        channels=3):
+    """Low-SNR structured synthetic images — the algorithmic quality gate.
+
+    The separable `make_synthetic` saturates any model at ~100% (it proves
+    plumbing, not algorithm quality). Here each class is a random spatial
+    TEMPLATE (smooth low-frequency pattern) mixed at low amplitude into
+    per-image noise plus a shared distractor pattern, so: accuracy stays
+    well below 100%, a 2000-exemplar replay budget is genuinely scarce, and
+    catastrophic forgetting / weight-align / KD effects become measurable
+    (tools/ablation.py). Templates depend only on (num_classes, size) so
+    train/eval share them; `seed` varies noise."""
+    trng = np.random.default_rng(5678 + num_classes * 31 + size)
+    # classes are MIXTURES of a small shared pattern basis (so classes
+    # genuinely overlap), spatially shifted per image (so pixel averaging
+    # does not trivially denoise them)
+    nbasis = 6
+    coarse = trng.normal(0, 1, size=(nbasis, 4, 4, channels))
+    reps = size // 4
+    basis = np.repeat(np.repeat(coarse, reps, axis=1), reps, axis=2)
+    coeff = trng.normal(0, 1, size=(num_classes, nbasis))
+    coeff /= np.linalg.norm(coeff, axis=1, keepdims=True)
+    templates = np.einsum("cb,bhwk->chwk", coeff, basis)
+    distractor = np.repeat(np.repeat(
+        trng.normal(0, 1, size=(8, 4, 4, channels)), reps, axis=1),
+        reps, axis=2)
+    rng = np.random.default_rng(seed)
+    xs, ys = [], []
+    for c in range(num_classes):
+        n = rng.normal(0, 1, size=(per_class, size, size, channels))
+        d = distractor[rng.integers(0, 8, per_class)]
+        amp = rng.uniform(0.6, 1.4, size=(per_class, 1, 1, 1))
+        t = np.broadcast_to(templates[c], (per_class, size, size, channels))
+        # per-image random cyclic shift of the class pattern
+        sh = rng.integers(0, size, size=(per_class, 2))
+        t = np.stack([np.roll(t[i], (sh[i, 0], sh[i, 1]), axis=(0, 1))
+                      for i in range(per_class)])
+        img = 128 + 30 * (0.40 * amp * t + 0.7 * d + 1.0 * n)
+        xs.append(np.clip(img, 0, 255).astype(np.uint8))
+        ys.append(np.full(per_class, c, dtype=np.int64))
+    x = np.concatenate(xs)
+    y = np.concatenate(ys)
+    perm = rng.permutation(len(y))
+    return x[perm], y[perm]
+
+
 def load_image(path):
     """Decode one image file -> uint8 HWC RGB (lazy path-based datasets)."""
     with Image.open(path) as im:
@@ -102,6 +148,13 @@ def build_source(args, is_train):
         per_class = max(args.synthetic_train_size // nc, 4) if is_train else 10
         x, y = make_synthetic(nc, per_class, args.input_size,
                               seed=0 if is_train else 1)
+        return x, y, nc, "synthetic"
+    if name == "synthetic_hard":
+        nc = getattr(args, "synthetic_classes", 100)
+        per_class = max(args.synthetic_train_size // nc, 4) if is_train \
+            else 40
+        x, y = make_synthetic_hard(nc, per_class, args.input_size,
+                                   seed=0 if is_train else 1)
         return x, y, nc, "synthetic"
     if name in ("imagenet100", "imagenet1000", "cub200"):
         nc = {"imagenet100": 100, "imagenet1000": 1000, "cub200": 200}[name]

@@ -371,7 +371,9 @@ def _run_tasks(args, device, watchdog):
         use_device_replay = (use_gpu_data
                              and not getattr(args, "no_device_replay", False))
         extra = None
-        if task_id > 0:
+        if task_id > 0 and getattr(args, "no_replay", False):
+            pass  # ablation: train each task without rehearsal
+        elif task_id > 0:
             if use_device_replay:
                 # replay straight from the HBM-resident mirror — exemplars
                 # never round-trip through the host (replaces the reference's
@@ -440,7 +442,8 @@ def _run_tasks(args, device, watchdog):
         train_one_task(model, teacher, engine, optimizer, scheduler,
                        train_loader, train_sampler, val_loader, device, args)
 
-        model.after_model_adaption(args.increment_per_task, args)
+        if not getattr(args, "no_wa", False):
+            model.after_model_adaption(args.increment_per_task, args)
         acc1 = evaluate(model, val_loader, device, args)
         acc1s.append(acc1)
         print(f"task id = {task_id}  @Acc1 = {acc1:.5f}, acc1s = {acc1s}")
