@@ -29,6 +29,8 @@ struct ConvGeom {
   // q = (n * m) >> 40 == n / d. Runtime-divisor division otherwise expands
   // to a ~40-instruction v_rcp sequence inside the hot loop.
   unsigned long long m_howo, m_wo;
+  // slow-path (tiny-C / ragged) per-element decode divisors: k/(C or K), rs/S
+  unsigned long long m_ck, m_s;
 };
 
 static inline unsigned long long magic40(int d) {
@@ -196,15 +198,19 @@ void conv2d_fwd_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
           for (int j = 0; j < 16; ++j) {
             int k = k0 + q * 32 + ahalf * 16 + j;
-            bf16_t v = 0;
-            if (arow_ok[rr] && k < CRS) {
-              int rs = k / g.C, c = k - rs * g.C;
-              int r = rs / g.S, s = rs - r * g.S;
-              int hi = ahb[rr] + r, wi = awb[rr] + s;
-              if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-                v = x[(((long)an[rr] * g.H + hi) * g.W + wi) * g.C + c];
-            }
-            tmp[j] = v;
+            int kc = min(k, CRS - 1);
+            int rs = (int)mdiv40((unsigned)kc, g.m_ck);
+            int c = kc - rs * g.C;
+            int r = (int)mdiv40((unsigned)rs, g.m_s);
+            int s = rs - r * g.S;
+            int hi = ahb[rr] + r, wi = awb[rr] + s;
+            bool ok = arow_ok[rr] & (k < CRS)
+                      & ((unsigned)hi < (unsigned)g.H)
+                      & ((unsigned)wi < (unsigned)g.W);
+            int hic = min(max(hi, 0), g.H - 1);
+            int wic = min(max(wi, 0), g.W - 1);
+            bf16_t v = x[(((long)an[rr] * g.H + hic) * g.W + wic) * g.C + c];
+            tmp[j] = ok ? v : (bf16_t)0;
           }
           areg[rr][2 * q] = *(int4*)&tmp[0];
           areg[rr][2 * q + 1] = *(int4*)&tmp[8];
@@ -396,19 +402,22 @@ void conv2d_bwd_data_kernel(const bf16_t* __restrict__ dy,
 #pragma unroll
         for (int j = 0; j < 16; ++j) {
           int k = k0 + q * 32 + ahalf * 16 + j;
-          bf16_t v = 0;
-          if (arow_ok && k < RSK) {
-            int rs = k / g.K, kc = k - rs * g.K;
-            int r = rs / g.S, s = rs - r * g.S;
-            int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
-            if (ho2 >= 0 && wo2 >= 0 && (ho2 % g.stride) == 0 &&
-                (wo2 % g.stride) == 0) {
-              int ho = ho2 / g.stride, wo = wo2 / g.stride;
-              if (ho < g.Ho && wo < g.Wo)
-                v = dy[(((long)an * g.Ho + ho) * g.Wo + wo) * g.K + kc];
-            }
-          }
-          tmp[j] = v;
+          int kk = min(k, RSK - 1);
+          int rs = (int)mdiv40((unsigned)kk, g.m_ck);
+          int kc = kk - rs * g.K;
+          int r = (int)mdiv40((unsigned)rs, g.m_s);
+          int s = rs - r * g.S;
+          int ho2 = ahi + g.pad - r, wo2 = awi + g.pad2 - s;
+          const bool s2d = g.stride == 2;
+          int ho = s2d ? (ho2 >> 1) : ho2;
+          int wo = s2d ? (wo2 >> 1) : wo2;
+          bool par = !s2d | (((ho2 | wo2) & 1) == 0);
+          bool ok = arow_ok & (k < RSK) & (ho2 >= 0) & (wo2 >= 0) & par
+                    & (ho < g.Ho) & (wo < g.Wo);
+          int hoc = min(max(ho, 0), g.Ho - 1);
+          int woc = min(max(wo, 0), g.Wo - 1);
+          bf16_t v = dy[(((long)an * g.Ho + hoc) * g.Wo + woc) * g.K + kc];
+          tmp[j] = ok ? v : (bf16_t)0;
         }
         areg[2 * q] = *(int4*)&tmp[0];
         areg[2 * q + 1] = *(int4*)&tmp[8];
@@ -776,16 +785,19 @@ void conv2d_bwd_weight_kernel(const bf16_t* __restrict__ dy,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int row = rowb + j;
-        bf16_t v = 0;
-        if (row < CRS && m_ok) {
-          int rs = row / g.C, c = row - rs * g.C;
-          int r = rs / g.S, s = rs - r * g.S;
-          int hi = ho * g.stride - g.pad + r;
-          int wi = wo * g.stride - g.pad2 + s;
-          if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
-            v = x[(((long)n * g.H + hi) * g.W + wi) * g.C + c];
-        }
-        adst[j] = v;
+        int rowc = min(row, CRS - 1);
+        int rs = (int)mdiv40((unsigned)rowc, g.m_ck);
+        int c = rowc - rs * g.C;
+        int r = (int)mdiv40((unsigned)rs, g.m_s);
+        int s = rs - r * g.S;
+        int hi = ho * g.stride - g.pad + r;
+        int wi = wo * g.stride - g.pad2 + s;
+        bool ok = (row < CRS) & m_ok & ((unsigned)hi < (unsigned)g.H)
+                  & ((unsigned)wi < (unsigned)g.W);
+        int hic = min(max(hi, 0), g.H - 1);
+        int wic = min(max(wi, 0), g.W - 1);
+        bf16_t v = x[(((long)n * g.H + hic) * g.W + wic) * g.C + c];
+        adst[j] = ok ? v : (bf16_t)0;
       }
     }
     if (ko0 + BN <= g.K) {  // block-uniform: full 64-col dY tile
@@ -970,7 +982,8 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
                       int N, int H, int W, int C, int K, int R, int S,
                       int stride, int pad, int Ho, int Wo, int ksplit,
                       void* stream) {
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
+             0, 0, magic40(C), magic40(S)};
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int fast_a = (C % 16 == 0);
@@ -1024,7 +1037,8 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
                            int N, int H, int W, int C, int K, int R, int S,
                            int stride, int pad, int Ho, int Wo, int ksplit,
                            void* stream) {
-  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
+             0, 0, magic40(K), magic40(S)};
   int H2max = (H + 1) >> 1, W2max = (W + 1) >> 1;
   int Mc = N * H2max * W2max;
   if (stride == 2 && (R > 1 || S > 1) && K % 16 == 0 &&
@@ -1080,7 +1094,8 @@ void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
                                int R, int S, int padh, int padw, int Ho,
                                int Wo, int ksplit, void* stream) {
   // stride-1 sub-problem of the parity decomposition (asymmetric pads)
-  ConvGeom g{N, H, W, C, K, R, S, 1, padh, padw, Ho, Wo};
+  ConvGeom g{N, H, W, C, K, R, S, 1, padh, padw, Ho, Wo,
+             0, 0, magic40(K), magic40(S)};
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);  // stride fixed at 1 here
@@ -1142,7 +1157,7 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
                              int Wo, int nslices, int accum, void* stream) {
   (void)mt;  // kept in the ABI for the (cached) im2col table experiments
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
-             magic40(Ho * Wo), magic40(Wo)};
+             magic40(Ho * Wo), magic40(Wo), magic40(C), magic40(S)};
   int M = N * Ho * Wo;
   if (M >= (1 << 22) || Ho * Wo >= (1 << 14)) {
     fprintf(stderr, "cilfw_conv2d_bwd_weight: magic-division range exceeded "

@@ -34,6 +34,10 @@ BASE = [
     "--synthetic_train_size", "50000", "--memory_size", "2000",
     "--eval_every_epoch", "0", "--input_size", "32",
     "--gpu_data", "--metric_every", "8", "--seed", "0",
+    # augmentation stays OFF: RandAugment obliterates the low-amplitude
+    # synthetic class signal (measured: all arms at chance with it on);
+    # the gate measures CIL dynamics, not augmentation robustness
+    "--no_aug",
 ]
 
 ARMS = {
