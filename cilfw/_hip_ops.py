@@ -223,8 +223,13 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S, out=None, accum=False):
     _bf16(x, "conv2d_bwd_weight.x")
     N, H, W_, C = x.shape
     _, Ho, Wo, K = dy.shape
-    if C < 16 and R * S > 1 and K % 16 == 0 \
-            and os.environ.get("CILFW_STEM_IM2COL") == "1":
+    # 7x7 stems (ImageNet): the per-element gather path measured 542 us on
+    # the 224^2 stem dW; padded-im2col + flat fast-path GEMM is far cheaper
+    # there (3x3 CIFAR stems measured better on the direct gather)
+    use_im2col = (os.environ.get("CILFW_STEM_IM2COL") == "1"
+                  or (R * S >= 25
+                      and os.environ.get("CILFW_STEM_IM2COL") != "0"))
+    if C < 16 and R * S > 1 and K % 16 == 0 and use_im2col:
         CRS = C * R * S
         CRSpad = (CRS + 15) // 16 * 16
         col, _, _ = _stem_cols(x, stride, pad, R, S, CRSpad)

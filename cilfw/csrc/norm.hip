@@ -149,7 +149,11 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
   }
 }
 
-// pass 3: y = gamma*(x-mean)*invstd + beta (+relu). 8 bf16 per thread.
+// pass 3: y = x*sc + sh (+res, +relu), sc = gamma*invstd,
+// sh = beta - mean*gamma*invstd. Folding BN to scale/shift lets the per-
+// thread channel params come from LDS as FOUR ds_read_b128 instead of 32
+// scalar reads (the scalar form measured LDS-bound at ~2.4 TB/s on the
+// ResNet-50 stem shapes — less than half the HBM roofline).
 __global__ __launch_bounds__(NT)
 void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
                      const bf16_t* __restrict__ res,
@@ -158,20 +162,18 @@ void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
                      const float* __restrict__ mean,
                      const float* __restrict__ invstd,
                      long total, int C, int relu) {
-  extern __shared__ float params[];  // [4][C]: gamma,beta,mean,invstd
+  extern __shared__ float params[];  // [2][C]: scale, shift
   for (int c = threadIdx.x; c < C; c += NT) {
-    params[c] = gamma[c];
-    params[C + c] = beta[c];
-    params[2 * C + c] = mean[c];
-    params[3 * C + c] = invstd[c];
+    float sc = gamma[c] * invstd[c];
+    params[c] = sc;
+    params[C + c] = beta[c] - mean[c] * sc;
   }
   __syncthreads();
   long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
   if (i0 + 8 > total) {
     for (long i = i0; i < total; ++i) {
       int c = (int)(i % C);
-      float v = (bf2f(x[i]) - params[2 * C + c]) * params[3 * C + c] *
-                params[c] + params[C + c];
+      float v = bf2f(x[i]) * params[c] + params[C + c];
       if (res != nullptr) v += bf2f(res[i]);
       if (relu) v = fmaxf(v, 0.f);
       y[i] = f2bf(v);
@@ -184,12 +186,16 @@ void bn_apply_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
   const bf16_t* re = (const bf16_t*)&rv;
   if (res != nullptr) rv = *(const int4*)&res[i0];
   bf16_t out[8];
-  int c0 = (int)(i0 % C);  // C % 8 == 0 for all cilfw models
+  int c0 = (int)(i0 % C);  // C % 8 == 0 for all cilfw models (8-aligned)
+  float4 sc0 = *(const float4*)&params[c0];
+  float4 sc1 = *(const float4*)&params[c0 + 4];
+  float4 sh0 = *(const float4*)&params[C + c0];
+  float4 sh1 = *(const float4*)&params[C + c0 + 4];
+  float scv[8] = {sc0.x, sc0.y, sc0.z, sc0.w, sc1.x, sc1.y, sc1.z, sc1.w};
+  float shv[8] = {sh0.x, sh0.y, sh0.z, sh0.w, sh1.x, sh1.y, sh1.z, sh1.w};
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    int c = c0 + j;
-    float v = (bf2f(xe[j]) - params[2 * C + c]) * params[3 * C + c] *
-              params[c] + params[C + c];
+    float v = bf2f(xe[j]) * scv[j] + shv[j];
     if (res != nullptr) v += bf2f(re[j]);
     if (relu) v = fmaxf(v, 0.f);
     out[j] = f2bf(v);
@@ -287,16 +293,24 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
                          const float* __restrict__ dgamma,
                          const float* __restrict__ dbeta,
                          long total, long M, int C, int relu, int training) {
-  extern __shared__ float params[];  // [5][C]: gamma*invstd, mean, invstd, dg, db
+  // folded: dx = P*dy' + Q*x + R  (P = gamma*invstd; training adds the
+  // batch-stat correction terms; eval has Q = R = 0) — params come from LDS
+  // as float4 pairs like bn_apply (the 5-array scalar form was LDS-bound)
+  extern __shared__ float params[];  // [3][C]: P, Q, R
+  const float rM = 1.f / (float)M;
   for (int c = threadIdx.x; c < C; c += NT) {
-    params[c] = gamma[c] * invstd[c];
-    params[C + c] = mean[c];
-    params[2 * C + c] = invstd[c];
-    params[3 * C + c] = dgamma[c];
-    params[4 * C + c] = dbeta[c];
+    float P = gamma[c] * invstd[c];
+    float Q = 0.f, R = 0.f;
+    if (training) {
+      float A = P * rM;
+      Q = -A * dgamma[c] * invstd[c];
+      R = A * (dgamma[c] * invstd[c] * mean[c] - dbeta[c]);
+    }
+    params[c] = P;
+    params[C + c] = Q;
+    params[2 * C + c] = R;
   }
   __syncthreads();
-  const float rM = 1.f / (float)M;
   long i0 = ((long)blockIdx.x * NT + threadIdx.x) * 8;
   if (i0 >= total) return;
   if (i0 + 8 <= total) {  // vectorized b128 path (C % 8 == 0)
@@ -310,20 +324,21 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
     __align__(16) bf16_t odx[8];
     __align__(16) bf16_t ores[8];
     int c0 = (int)(i0 % C);
+    float4 p0 = *(const float4*)&params[c0];
+    float4 p1 = *(const float4*)&params[c0 + 4];
+    float4 q0 = *(const float4*)&params[C + c0];
+    float4 q1 = *(const float4*)&params[C + c0 + 4];
+    float4 r0 = *(const float4*)&params[2 * C + c0];
+    float4 r1 = *(const float4*)&params[2 * C + c0 + 4];
+    float pv[8] = {p0.x, p0.y, p0.z, p0.w, p1.x, p1.y, p1.z, p1.w};
+    float qv[8] = {q0.x, q0.y, q0.z, q0.w, q1.x, q1.y, q1.z, q1.w};
+    float rv[8] = {r0.x, r0.y, r0.z, r0.w, r1.x, r1.y, r1.z, r1.w};
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int c = c0 + j;
       float g = bf2f(de[j]);
       if (relu && bf2f(ye[j]) <= 0.f) g = 0.f;
       ores[j] = f2bf(g);
-      float v;
-      if (training) {
-        float xhat = (bf2f(xe[j]) - params[C + c]) * params[2 * C + c];
-        v = params[c] * rM *
-            ((float)M * g - params[4 * C + c] - xhat * params[3 * C + c]);
-      } else {
-        v = g * params[c];
-      }
+      float v = pv[j] * g + qv[j] * bf2f(xe[j]) + rv[j];
       odx[j] = f2bf(v);
     }
     *(int4*)&dx[i0] = *(int4*)odx;
@@ -335,15 +350,8 @@ void bn_bwd_apply_kernel(const bf16_t* __restrict__ dy,
     float g = bf2f(dy[i]);
     if (relu && bf2f(y[i]) <= 0.f) g = 0.f;
     if (dres != nullptr) dres[i] = f2bf(g);
-    float v;
-    if (training) {
-      float xhat = (bf2f(x[i]) - params[C + c]) * params[2 * C + c];
-      v = params[c] * rM *
-          ((float)M * g - params[4 * C + c] - xhat * params[3 * C + c]);
-    } else {
-      v = g * params[c];
-    }
-    dx[i] = f2bf(v);
+    dx[i] = f2bf(params[c] * g + params[C + c] * bf2f(x[i])
+                 + params[2 * C + c]);
   }
 }
 
@@ -583,7 +591,7 @@ void cilfw_bn_fwd(const void* x, void* y, const void* res,
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_apply_kernel, dim3((int)blocks), dim3(NT),
-                     4 * C * sizeof(float), st, (const bf16_t*)x, (bf16_t*)y,
+                     2 * C * sizeof(float), st, (const bf16_t*)x, (bf16_t*)y,
                      (const bf16_t*)res, (const float*)gamma,
                      (const float*)beta, (const float*)mean,
                      (const float*)invstd, total, C, relu);
@@ -596,7 +604,7 @@ void cilfw_bn_apply_only(const void* x, void* y, const void* res,
   // frozen-model eval: stats precomputed, apply is the only launch
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_apply_kernel, dim3((int)blocks), dim3(NT),
-                     4 * C * sizeof(float), (hipStream_t)stream,
+                     2 * C * sizeof(float), (hipStream_t)stream,
                      (const bf16_t*)x, (bf16_t*)y, (const bf16_t*)res,
                      (const float*)gamma, (const float*)beta,
                      (const float*)mean, (const float*)invstd, total, C,
@@ -626,7 +634,7 @@ void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);
   hipLaunchKernelGGL(bn_bwd_apply_kernel, dim3((int)blocks), dim3(NT),
-                     5 * C * sizeof(float), st, (const bf16_t*)dy,
+                     3 * C * sizeof(float), st, (const bf16_t*)dy,
                      (const bf16_t*)x, (const bf16_t*)y, (bf16_t*)dx,
                      (bf16_t*)dres, (const float*)gamma, (const float*)mean,
                      (const float*)invstd, (const float*)dgamma,
