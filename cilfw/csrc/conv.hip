@@ -959,6 +959,15 @@ static int bk64_min_crs() {
   return v;
 }
 
+static int bk64_max_blocks() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_BK64_MAXBLOCKS");
+    v = e ? atoi(e) : 768;  // above this, BKT=32's higher occupancy wins
+  }
+  return v;
+}
+
 static int pick_ksplit(int nblocks, int nk) {
   // aim for >= 512 workgroups (2 per CU) without shredding the K loop
   int ks = 1;
@@ -993,7 +1002,7 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   // (a BN=128 tile variant was tried and measured slower AND failed numerics
   //  — removed; the template's NRC generalization remains at the validated 2)
   int use64 = bm == BM && (CRS >= bk64_min_crs()) &&
-              (cdiv(M, BM) * cdiv(K, BN) * ksplit < 768);
+              (cdiv(M, BM) * cdiv(K, BN) * ksplit < bk64_max_blocks());
   int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, bm), cdiv(K, BN), ksplit);
 #define LAUNCH_FWD(BKT_, BM_)                                                 \
@@ -1057,7 +1066,7 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0) && stride <= 2;  // FAST path shift-divides
   int use64 = (RSK >= bk64_min_crs()) &&
-              (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
+              (cdiv(M, BM) * cdiv(C, BN) * ksplit < bk64_max_blocks());
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
   if (use64 && fast_a)
@@ -1100,7 +1109,7 @@ void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);  // stride fixed at 1 here
   int use64 = (RSK >= bk64_min_crs()) &&
-              (cdiv(M, BM) * cdiv(C, BN) * ksplit < 768);
+              (cdiv(M, BM) * cdiv(C, BN) * ksplit < bk64_max_blocks());
   int nk = cdiv(RSK, use64 ? 64 : 32);
   dim3 grid(cdiv(M, BM), cdiv(C, BN), ksplit);
   if (use64 && fast_a)
