@@ -942,6 +942,178 @@ void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
   mt[m] = (n << 20) | ((ho * stride) << 10) | (wo * stride);
 }
 
+
+// ========================= forward v2 (all-glds) =========================
+// 2-phase all-global_load_lds staging (guide T3 minimum 2-phase + the glds
+// rows of the staging table): the A tile's gather sources are PER-LANE glds
+// addresses (a zero page absorbs OOB lanes), XOR-swizzled at the source so
+// the linear LDS image reads conflict-free; the B tile lands as [ngrp][k][16]
+// images consumed by ds_read_b64_tr_b16 pairs (w is k-major — glds cannot
+// transpose). One raw barrier + vmcnt(0) per K-step, whole-phase setprio.
+// Requires C % 8 == 0 (an 8-elem chunk stays inside one (r,s)).
+// Measured (tools/probe/convfwd_v2.hip): l1 337 TF, l2 375, l3 257 vs the
+// register-staged v1's 274/261/219; 3-buffer counted-vmcnt was null at this
+// occupancy (3 blocks/CU), BM=256 and BKT=64-everywhere lose to occupancy.
+
+#define V2BK 64
+#define V2_A_ELEMS (BM * V2BK)
+#define V2_B_ELEMS (4 * V2BK * 16)
+#define V2_BUF (V2_A_ELEMS + V2_B_ELEMS)
+
+__global__ __launch_bounds__(NTHREADS)
+void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
+                          const bf16_t* __restrict__ w,
+                          bf16_t* __restrict__ y, float* __restrict__ ws,
+                          const bf16_t* __restrict__ zpage, ConvGeom g,
+                          int M, int CRS, int nk, int ksplit) {
+  __shared__ bf16_t lds[2 * V2_BUF];
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const int t = threadIdx.x;
+  const int wv = t >> 6, lane = t & 63;
+  const int wr = wv >> 1, wc = wv & 1;
+  const int steps = (nk + ksplit - 1) / ksplit;
+  const int kt0 = blockIdx.z * steps;
+  const int kt1 = min(kt0 + steps, nk);
+  const int nsteps = kt1 - kt0;
+
+  // A staging: 16 glds x 8 rows; wave wv does instrs wv*4..+3. Lane covers
+  // row wv*32 + i*8 + l/8, source k-chunk (l%8) ^ (row&7) (source swizzle).
+  int achunk[4], ar[4], as_[4], ac[4];
+  bool mok[4];
+  int an[4], ahb[4], awb[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wv * 32 + i * 8 + (lane >> 3);
+    achunk[i] = (lane & 7) ^ (row & 7);
+    int m = m0 + row;
+    mok[i] = m < M;
+    int mc = min(m, M - 1);
+    int n = mc / (g.Ho * g.Wo);
+    int rem = mc - n * (g.Ho * g.Wo);
+    int aho = rem / g.Wo;
+    an[i] = n;
+    ahb[i] = aho * g.stride - g.pad;
+    awb[i] = (rem - aho * g.Wo) * g.stride - g.pad2;
+    int k = kt0 * V2BK + achunk[i] * 8;
+    int rs = k / g.C;
+    ac[i] = k - rs * g.C;
+    ar[i] = rs / g.S;
+    as_[i] = rs - ar[i] * g.S;
+  }
+  // B staging: 2 glds per wave into [ngrp=wv][k][16]
+  const int bk0 = lane >> 1;
+  const int bns = (lane & 1) * 8;
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    acc[i][0] = f32x4{0, 0, 0, 0};
+    acc[i][1] = f32x4{0, 0, 0, 0};
+  }
+
+  auto stage = [&](int buf, int kt) {
+    bf16_t* base = &lds[buf * V2_BUF];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int hi = ahb[i] + ar[i], wi = awb[i] + as_[i];
+      int k = kt * V2BK + achunk[i] * 8;
+      bool ok = mok[i] & (k < CRS) & ((unsigned)hi < (unsigned)g.H)
+                & ((unsigned)wi < (unsigned)g.W);
+      const bf16_t* src = ok
+          ? &x[(((long)an[i] * g.H + hi) * g.W + wi) * g.C + ac[i]]
+          : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[(wv * 4 + i) * 8 * V2BK], 16, 0, 0);
+      int c = ac[i] + V2BK;
+      int r = ar[i], s = as_[i];
+      while (c >= g.C) { c -= g.C; if (++s == g.S) { s = 0; ++r; } }
+      ac[i] = c; ar[i] = r; as_[i] = s;
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int k = kt * V2BK + j * 32 + bk0;
+      int kc = min(k, CRS - 1);
+      bool ok = (k < CRS) & (n0 + wv * 16 + bns + 8 <= g.K);
+      const bf16_t* src = ok ? &w[(long)kc * g.K + n0 + wv * 16 + bns]
+                             : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[V2_A_ELEMS + wv * (V2BK * 16) + j * 32 * 16], 16, 0, 0);
+    }
+  };
+
+  const int fh = lane & 15, fq = lane >> 4;
+  const unsigned lds0 = (unsigned)(uintptr_t)&lds[0];
+  const unsigned btr_e = (unsigned)((fq * 8 + ((lane >> 2) & 3)) * 16
+                                    + (lane & 3) * 4);
+
+  if (nsteps > 0) stage(0, kt0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  for (int kt = kt0; kt < kt1; ++kt) {
+    int cur = (kt - kt0) & 1;
+    if (kt + 1 < kt1) stage(cur ^ 1, kt + 1);
+    const bf16_t* As = &lds[cur * V2_BUF];
+    const unsigned bbase = lds0 + 2u * (cur * V2_BUF + V2_A_ELEMS);
+    bf16x8 bfr[2][2];
+#pragma unroll
+    for (int q = 0; q < 2; ++q)
+#pragma unroll
+      for (int nr = 0; nr < 2; ++nr) {
+        unsigned a = bbase + 2u * ((wc * 2 + nr) * (V2BK * 16))
+                     + 2u * (btr_e + q * 32 * 16);
+        typedef __attribute__((ext_vector_type(4))) short s4_t;
+        s4_t lo, hi;
+        asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                     "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+                     : "=&v"(lo), "=&v"(hi) : "v"(a));
+        bfr[q][nr] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int kb = fq * 8 + q * 32;
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr) {
+        int row = wr * 64 + mr * 16 + fh;
+        int kcol = kb ^ ((row & 7) << 3);
+        bf16x8 afr = *(const bf16x8*)&As[row * V2BK + kcol];
+        if (q == 0 && mr == 0) {
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+          __builtin_amdgcn_sched_barrier(0);
+          __builtin_amdgcn_s_setprio(1);
+        }
+        acc[mr][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr, bfr[q][0], acc[mr][0], 0, 0, 0);
+        acc[mr][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr, bfr[q][1], acc[mr][1], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 2; ++nr)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mr * 16 + fq * 4 + r;
+        int col = n0 + wc * 32 + nr * 16 + fh;
+        if (row < M && col < g.K) {
+          if (ksplit > 1)
+            ws[((long)blockIdx.z * M + row) * g.K + col] = acc[mr][nr][r];
+          else
+            y[(long)row * g.K + col] = f2bf(acc[mr][nr][r]);
+        }
+      }
+}
+
 // ============================== launchers ==============================
 
 #include <stdlib.h>
@@ -977,6 +1149,32 @@ static int pick_ksplit(int nblocks, int nk) {
 
 extern "C" {
 
+// 64 KiB zeroed device page: OOB gather lanes of the all-glds v2 kernels
+// point here. Allocated lazily OUTSIDE graph capture (the engine always
+// runs an eager warmup step before capturing).
+static void* zpage_ptr() {
+  static void* p = nullptr;
+  if (p == nullptr) {
+    hipStreamCaptureStatus st = hipStreamCaptureStatusNone;
+    (void)hipStreamIsCapturing(nullptr, &st);
+    if (st != hipStreamCaptureStatusNone) return nullptr;  // caller falls back
+    void* q = nullptr;
+    if (hipMalloc(&q, 65536) != hipSuccess) return nullptr;
+    (void)hipMemset(q, 0, 65536);
+    p = q;
+  }
+  return p;
+}
+
+static int conv_v2_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_V2");
+    v = e ? atoi(e) : 1;
+  }
+  return v;
+}
+
 static int fwd_bm256_min_m() {
   static int v = -1;
   if (v < 0) {
@@ -1003,6 +1201,25 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
   //  — removed; the template's NRC generalization remains at the validated 2)
   int use64 = bm == BM && (CRS >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(K, BN) * ksplit < bk64_max_blocks());
+  void* zp = (C % 8 == 0 && conv_v2_enabled()) ? zpage_ptr() : nullptr;
+  if (zp != nullptr && bm == BM) {
+    // all-glds 2-phase kernel (v2): BK=64, tr_b16 B operand
+    int nk2 = cdiv(CRS, V2BK);
+    int ks2 = ksplit;
+    dim3 grid2(cdiv(M, BM), cdiv(K, BN), ks2);
+    hipLaunchKernelGGL(conv2d_fwd_v2_kernel, grid2, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)x,
+                       (const bf16_t*)w, (bf16_t*)y, (float*)ws,
+                       (const bf16_t*)zp, g, M, CRS, nk2, ks2);
+    if (ks2 > 1) {
+      long len = (long)M * K;
+      hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                         dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                         dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                         (bf16_t*)y, ks2, len);
+    }
+    return;
+  }
   int nk = cdiv(CRS, use64 ? 64 : 32);
   dim3 grid(cdiv(M, bm), cdiv(K, BN), ksplit);
 #define LAUNCH_FWD(BKT_, BM_)                                                 \
