@@ -1114,6 +1114,169 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
       }
 }
 
+
+// ==================== backward data v2 (all-glds) ====================
+// Same 2-phase all-glds structure as fwd v2. A = scattered dY gather
+// (stride handled by shift-select + parity mask); B = W consumed as
+// [c][rsk] — for a fixed channel c, consecutive kc are CONTIGUOUS in the
+// (R,S,C,K) weight layout, so B stages through plain per-lane glds with
+// the same XOR source swizzle as A and fragments read as b128 (no
+// transpose read needed, unlike fwd's B). Requires K % 8 == 0.
+
+__global__ __launch_bounds__(NTHREADS)
+void conv2d_bwd_data_v2_kernel(const bf16_t* __restrict__ dy,
+                               const bf16_t* __restrict__ w,
+                               bf16_t* __restrict__ dx,
+                               float* __restrict__ ws,
+                               const bf16_t* __restrict__ zpage, ConvGeom g,
+                               int M, int RSK, int nk, int ksplit) {
+  __shared__ bf16_t lds[2 * (BM * V2BK + BN * V2BK)];
+  const int A_E = BM * V2BK;
+  const int BUF = BM * V2BK + BN * V2BK;
+  const int m0 = blockIdx.x * BM;
+  const int c0 = blockIdx.y * BN;
+  const int t = threadIdx.x;
+  const int wv = t >> 6, lane = t & 63;
+  const int wr = wv >> 1, wc = wv & 1;
+  const int steps = (nk + ksplit - 1) / ksplit;
+  const int kt0 = blockIdx.z * steps;
+  const int kt1 = min(kt0 + steps, nk);
+
+  // A: 16 glds x 8 rows of the [128][64] dY-gather image
+  int achunk[4], ar[4], as_[4], akc[4];
+  bool mok[4];
+  int an[4], ahi[4], awi[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int row = wv * 32 + i * 8 + (lane >> 3);
+    achunk[i] = (lane & 7) ^ (row & 7);
+    int m = m0 + row;
+    mok[i] = m < M;
+    int mc = min(m, M - 1);
+    an[i] = mc / (g.H * g.W);
+    int rem = mc - an[i] * (g.H * g.W);
+    ahi[i] = rem / g.W;
+    awi[i] = rem - ahi[i] * g.W;
+    int k = kt0 * V2BK + achunk[i] * 8;
+    int rs = k / g.K;
+    akc[i] = k - rs * g.K;
+    ar[i] = rs / g.S;
+    as_[i] = rs - ar[i] * g.S;
+  }
+  // B: 8 glds (2/wave) into the [64 c][64 rsk] image, source swizzled
+  int bchunk[2], br[2], bs_[2], bkc[2], brow[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    brow[j] = wv * 16 + j * 8 + (lane >> 3);
+    bchunk[j] = (lane & 7) ^ (brow[j] & 7);
+    int k = kt0 * V2BK + bchunk[j] * 8;
+    int rs = k / g.K;
+    bkc[j] = k - rs * g.K;
+    br[j] = rs / g.S;
+    bs_[j] = rs - br[j] * g.S;
+  }
+
+  f32x4 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    acc[i][0] = f32x4{0, 0, 0, 0};
+    acc[i][1] = f32x4{0, 0, 0, 0};
+  }
+  const bool s2d = g.stride == 2;
+
+  auto stage = [&](int buf, int kt) {
+    bf16_t* base = &lds[buf * BUF];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int k = kt * V2BK + achunk[i] * 8;
+      int ho2 = ahi[i] + g.pad - ar[i], wo2 = awi[i] + g.pad2 - as_[i];
+      int ho = s2d ? (ho2 >> 1) : ho2;
+      int wo = s2d ? (wo2 >> 1) : wo2;
+      bool par = !s2d | (((ho2 | wo2) & 1) == 0);
+      bool ok = mok[i] & (k < RSK) & (ho2 >= 0) & (wo2 >= 0) & par
+                & (ho < g.Ho) & (wo < g.Wo);
+      const bf16_t* src = ok
+          ? &dy[(((long)an[i] * g.Ho + ho) * g.Wo + wo) * g.K + akc[i]]
+          : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[(wv * 4 + i) * 8 * V2BK], 16, 0, 0);
+      int kc = akc[i] + V2BK;
+      int r = ar[i], s = as_[i];
+      while (kc >= g.K) { kc -= g.K; if (++s == g.S) { s = 0; ++r; } }
+      akc[i] = kc; ar[i] = r; as_[i] = s;
+    }
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int k = kt * V2BK + bchunk[j] * 8;
+      int rs_orig = br[j] * g.S + bs_[j];
+      bool ok = (k < RSK) & (c0 + brow[j] < g.C) & (rs_orig < g.R * g.S);
+      const bf16_t* src = ok
+          ? &w[((long)rs_orig * g.C + c0 + brow[j]) * g.K + bkc[j]]
+          : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[A_E + (wv * 2 + j) * 8 * V2BK], 16, 0, 0);
+      int kc = bkc[j] + V2BK;
+      int r = br[j], s = bs_[j];
+      while (kc >= g.K) { kc -= g.K; if (++s == g.S) { s = 0; ++r; } }
+      bkc[j] = kc; br[j] = r; bs_[j] = s;
+    }
+  };
+
+  const int fh = lane & 15, fq = lane >> 4;
+
+  if (kt1 > kt0) stage(0, kt0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  for (int kt = kt0; kt < kt1; ++kt) {
+    int cur = (kt - kt0) & 1;
+    if (kt + 1 < kt1) stage(cur ^ 1, kt + 1);
+    const bf16_t* As = &lds[cur * BUF];
+    const bf16_t* Bs = &lds[cur * BUF + A_E];
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      const int kb = fq * 8 + q * 32;
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr) {
+        int row = wr * 64 + mr * 16 + fh;
+        bf16x8 afr = *(const bf16x8*)&As[row * V2BK
+                                         + (kb ^ ((row & 7) << 3))];
+        if (q == 0 && mr == 0) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int nr = 0; nr < 2; ++nr) {
+          int col = wc * 32 + nr * 16 + fh;
+          bf16x8 bfr = *(const bf16x8*)&Bs[col * V2BK
+                                           + (kb ^ ((col & 7) << 3))];
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr, acc[mr][nr], 0, 0, 0);
+        }
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+#pragma unroll
+  for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 2; ++nr)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + mr * 16 + fq * 4 + r;
+        int col = c0 + wc * 32 + nr * 16 + fh;
+        if (row < M && col < g.C) {
+          if (ksplit > 1)
+            ws[((long)blockIdx.z * M + row) * g.C + col] = acc[mr][nr][r];
+          else
+            dx[(long)row * g.C + col] = f2bf(acc[mr][nr][r]);
+        }
+      }
+}
+
 // ============================== launchers ==============================
 
 #include <stdlib.h>
@@ -1171,6 +1334,19 @@ static int conv_v2_enabled() {
   if (v < 0) {
     const char* e = getenv("CILFW_CONV_V2");
     v = e ? atoi(e) : 1;
+  }
+  return v;
+}
+
+// bwd-data v2 measured slower than v1 on the CIFAR shapes (l1 pair 129 vs
+// 126 us, l3 139 vs 126) but ahead on the large-M ImageNet layers; route by
+// M with an env override (CILFW_CONV_V2_BWD_MINM, 0 disables).
+static int conv_v2_bwd_minm() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("CILFW_CONV_V2_BWD_MINM");
+    v = e ? atoi(e) : 150000;
+    if (v == 0) v = 1 << 30;
   }
   return v;
 }
@@ -1282,6 +1458,24 @@ void cilfw_conv2d_bwd_data(const void* dy, const void* w, void* dx, void* ws,
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0) && stride <= 2;  // FAST path shift-divides
+  void* zp = (K % 8 == 0 && stride <= 2 && conv_v2_enabled()
+              && M >= conv_v2_bwd_minm()) ? zpage_ptr() : nullptr;
+  if (zp != nullptr) {  // all-glds 2-phase kernel (v2)
+    int nk2 = cdiv(RSK, V2BK);
+    dim3 grid2(cdiv(M, BM), cdiv(C, BN), ksplit);
+    hipLaunchKernelGGL(conv2d_bwd_data_v2_kernel, grid2, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws,
+                       (const bf16_t*)zp, g, M, RSK, nk2, ksplit);
+    if (ksplit > 1) {
+      long len = (long)M * C;
+      hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                         dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                         dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                         (bf16_t*)dx, ksplit, len);
+    }
+    return;
+  }
   int use64 = (RSK >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < bk64_max_blocks());
   int nk = cdiv(RSK, use64 ? 64 : 32);
@@ -1325,6 +1519,24 @@ void cilfw_conv2d_bwd_data_sub(const void* dy, const void* w, void* dx,
   int M = N * H * W;
   int RSK = R * S * K;
   int fast_a = (K % 16 == 0);  // stride fixed at 1 here
+  void* zp = (K % 8 == 0 && conv_v2_enabled()
+              && M >= conv_v2_bwd_minm()) ? zpage_ptr() : nullptr;
+  if (zp != nullptr) {  // all-glds 2-phase kernel (v2)
+    int nk2 = cdiv(RSK, V2BK);
+    dim3 grid2(cdiv(M, BM), cdiv(C, BN), ksplit);
+    hipLaunchKernelGGL(conv2d_bwd_data_v2_kernel, grid2, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)w, (bf16_t*)dx, (float*)ws,
+                       (const bf16_t*)zp, g, M, RSK, nk2, ksplit);
+    if (ksplit > 1) {
+      long len = (long)M * C;
+      hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                         dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                         dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                         (bf16_t*)dx, ksplit, len);
+    }
+    return;
+  }
   int use64 = (RSK >= bk64_min_crs()) &&
               (cdiv(M, BM) * cdiv(C, BN) * ksplit < bk64_max_blocks());
   int nk = cdiv(RSK, use64 ? 64 : 32);
