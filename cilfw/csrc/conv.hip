@@ -960,15 +960,20 @@ void fill_mtable_kernel(int* __restrict__ mt, int M, int HoWo, int Wo,
 #define V2_B_ELEMS (4 * V2BK * 16)
 #define V2_BUF (V2_A_ELEMS + V2_B_ELEMS)
 
+template <int BNX2 = BN>
 __global__ __launch_bounds__(NTHREADS)
 void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
                           const bf16_t* __restrict__ w,
                           bf16_t* __restrict__ y, float* __restrict__ ws,
                           const bf16_t* __restrict__ zpage, ConvGeom g,
                           int M, int CRS, int nk, int ksplit) {
-  __shared__ bf16_t lds[2 * V2_BUF];
+  constexpr int NGRP = BNX2 / 16;       // 16-col groups in the B tile
+  constexpr int NRC = BNX2 / 32;        // N-fragments per wave
+  constexpr int BEL = NGRP * V2BK * 16; // B image elems
+  constexpr int BUFE = V2_A_ELEMS + BEL;
+  __shared__ bf16_t lds[2 * BUFE];
   const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  const int n0 = blockIdx.y * BNX2;
   const int t = threadIdx.x;
   const int wv = t >> 6, lane = t & 63;
   const int wr = wv >> 1, wc = wv & 1;
@@ -1005,15 +1010,14 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
   const int bk0 = lane >> 1;
   const int bns = (lane & 1) * 8;
 
-  f32x4 acc[4][2];
+  f32x4 acc[4][NRC];
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    acc[i][0] = f32x4{0, 0, 0, 0};
-    acc[i][1] = f32x4{0, 0, 0, 0};
-  }
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < NRC; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
   auto stage = [&](int buf, int kt) {
-    bf16_t* base = &lds[buf * V2_BUF];
+    bf16_t* base = &lds[buf * BUFE];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       int hi = ahb[i] + ar[i], wi = awb[i] + as_[i];
@@ -1032,17 +1036,23 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
       while (c >= g.C) { c -= g.C; if (++s == g.S) { s = 0; ++r; } }
       ac[i] = c; ar[i] = r; as_[i] = s;
     }
+    // B: (NGRP/4) groups per wave x 2 glds each
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      int k = kt * V2BK + j * 32 + bk0;
-      int kc = min(k, CRS - 1);
-      bool ok = (k < CRS) & (n0 + wv * 16 + bns + 8 <= g.K);
-      const bf16_t* src = ok ? &w[(long)kc * g.K + n0 + wv * 16 + bns]
-                             : zpage;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)src,
-          (__attribute__((address_space(3))) unsigned int*)
-              &base[V2_A_ELEMS + wv * (V2BK * 16) + j * 32 * 16], 16, 0, 0);
+    for (int gset = 0; gset < NGRP / 4; ++gset) {
+      const int ng = wv + gset * 4;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int k = kt * V2BK + j * 32 + bk0;
+        int kc = min(k, CRS - 1);
+        bool ok = (k < CRS) & (n0 + ng * 16 + bns + 8 <= g.K);
+        const bf16_t* src = ok ? &w[(long)kc * g.K + n0 + ng * 16 + bns]
+                               : zpage;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)src,
+            (__attribute__((address_space(3))) unsigned int*)
+                &base[V2_A_ELEMS + ng * (V2BK * 16) + j * 32 * 16],
+            16, 0, 0);
+      }
     }
   };
 
@@ -1057,14 +1067,14 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
   for (int kt = kt0; kt < kt1; ++kt) {
     int cur = (kt - kt0) & 1;
     if (kt + 1 < kt1) stage(cur ^ 1, kt + 1);
-    const bf16_t* As = &lds[cur * V2_BUF];
-    const unsigned bbase = lds0 + 2u * (cur * V2_BUF + V2_A_ELEMS);
-    bf16x8 bfr[2][2];
+    const bf16_t* As = &lds[cur * BUFE];
+    const unsigned bbase = lds0 + 2u * (cur * BUFE + V2_A_ELEMS);
+    bf16x8 bfr[2][NRC];
 #pragma unroll
     for (int q = 0; q < 2; ++q)
 #pragma unroll
-      for (int nr = 0; nr < 2; ++nr) {
-        unsigned a = bbase + 2u * ((wc * 2 + nr) * (V2BK * 16))
+      for (int nr = 0; nr < NRC; ++nr) {
+        unsigned a = bbase + 2u * ((wc * NRC + nr) * (V2BK * 16))
                      + 2u * (btr_e + q * 32 * 16);
         typedef __attribute__((ext_vector_type(4))) short s4_t;
         s4_t lo, hi;
@@ -1086,10 +1096,10 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
           __builtin_amdgcn_sched_barrier(0);
           __builtin_amdgcn_s_setprio(1);
         }
-        acc[mr][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afr, bfr[q][0], acc[mr][0], 0, 0, 0);
-        acc[mr][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afr, bfr[q][1], acc[mr][1], 0, 0, 0);
+#pragma unroll
+        for (int nr = 0; nr < NRC; ++nr)
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[q][nr], acc[mr][nr], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
@@ -1100,11 +1110,11 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
 #pragma unroll
   for (int mr = 0; mr < 4; ++mr)
 #pragma unroll
-    for (int nr = 0; nr < 2; ++nr)
+    for (int nr = 0; nr < NRC; ++nr)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + mr * 16 + fq * 4 + r;
-        int col = n0 + wc * 32 + nr * 16 + fh;
+        int col = n0 + wc * (16 * NRC) + nr * 16 + fh;
         if (row < M && col < g.K) {
           if (ksplit > 1)
             ws[((long)blockIdx.z * M + row) * g.K + col] = acc[mr][nr][r];
@@ -1379,11 +1389,36 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
               (cdiv(M, BM) * cdiv(K, BN) * ksplit < bk64_max_blocks());
   void* zp = (C % 8 == 0 && conv_v2_enabled()) ? zpage_ptr() : nullptr;
   if (zp != nullptr && bm == BM) {
-    // all-glds 2-phase kernel (v2): BK=64, tr_b16 B operand
+    // all-glds 2-phase kernel (v2): BK=64, tr_b16 B operand. A 128-wide
+    // N-tile (2x the MFMA per staged A-byte at 2 blocks/CU) for K >= 128
+    // when the grid still fills the chip; env CILFW_CONV_V2_BN128=0 off.
     int nk2 = cdiv(CRS, V2BK);
     int ks2 = ksplit;
+    static int bn128 = -1;
+    if (bn128 < 0) {
+      const char* e = getenv("CILFW_CONV_V2_BN128");
+      bn128 = e ? atoi(e) : 1;
+    }
+    // BN=128 wins only where the A-gather is trivial (1x1 convs: rn50_1x1
+    // 173->218 TF); on 3x3 layers the halved occupancy loses (l2 348->318)
+    if (bn128 && K % 128 == 0 && R == 1 && S == 1 &&
+        cdiv(M, BM) * cdiv(K, 128) * ks2 >= 256) {
+      dim3 grid2(cdiv(M, BM), cdiv(K, 128), ks2);
+      hipLaunchKernelGGL(conv2d_fwd_v2_kernel<128>, grid2, dim3(NTHREADS),
+                         0, (hipStream_t)stream, (const bf16_t*)x,
+                         (const bf16_t*)w, (bf16_t*)y, (float*)ws,
+                         (const bf16_t*)zp, g, M, CRS, nk2, ks2);
+      if (ks2 > 1) {
+        long len = (long)M * K;
+        hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                           dim3((int)cdiv((long)len, (long)NTHREADS * 4)),
+                           dim3(NTHREADS), 0, (hipStream_t)stream,
+                           (float*)ws, (bf16_t*)y, ks2, len);
+      }
+      return;
+    }
     dim3 grid2(cdiv(M, BM), cdiv(K, BN), ks2);
-    hipLaunchKernelGGL(conv2d_fwd_v2_kernel, grid2, dim3(NTHREADS), 0,
+    hipLaunchKernelGGL(conv2d_fwd_v2_kernel<BN>, grid2, dim3(NTHREADS), 0,
                        (hipStream_t)stream, (const bf16_t*)x,
                        (const bf16_t*)w, (bf16_t*)y, (float*)ws,
                        (const bf16_t*)zp, g, M, CRS, nk2, ks2);
