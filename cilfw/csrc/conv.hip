@@ -1287,6 +1287,167 @@ void conv2d_bwd_data_v2_kernel(const bf16_t* __restrict__ dy,
       }
 }
 
+
+// ==================== backward weight v2 (all-glds) ====================
+// dW[(r,s,c), k] = sum_m X-gather[m] * dY[m]: BOTH operands are m-major at
+// the source (x: 16 B = 8 CRS rows of one pixel; dy: 16 B = 8 K of one m),
+// exactly the [grp][m][16] tr_b16 image shape — so both stage by
+// global_load_lds (no register round-trip, no ds_write pass) and fragments
+// come back as ds_read_b64_tr_b16 pairs. 2-phase raw-barrier loop like
+// fwd v2. Requires C % 8 == 0 and K % 8 == 0.
+
+#define W2_IMG (4 * WBK * 16)   // one operand image: [4 grp][64 m][16]
+
+__global__ __launch_bounds__(NTHREADS)
+void conv2d_bwd_weight_v2_kernel(const bf16_t* __restrict__ dy,
+                                 const bf16_t* __restrict__ x,
+                                 float* __restrict__ ws,
+                                 const bf16_t* __restrict__ zpage,
+                                 ConvGeom g, int M, int CRS, int slice_len) {
+  __shared__ bf16_t lds[2 * 2 * W2_IMG];
+  const int rs0 = blockIdx.x * WBM;
+  const int ko0 = blockIdx.y * BN;
+  const int ms = blockIdx.z * slice_len;
+  const int me = min(ms + slice_len, M);
+  const int t = threadIdx.x;
+  const int wv = t >> 6, lane = t & 63;
+  const int wr = wv >> 1, wc = wv & 1;
+  const int HoWo = g.Ho * g.Wo;
+  const bool flat = (g.H == 1 && g.W == 1 && g.Ho == 1 && g.Wo == 1);
+
+  // lane's m within the 64-m tile and its 8-row chunk: 4 glds per wave per
+  // operand; instr i covers m rows [i*16, i*16+16), lane -> m = i*16 + l/4,
+  // chunk = l%4 is WRONG for 16-wide rows: [64 m][16] rows are 32 B = 2
+  // lanes -> lane covers m = i*32 + l/2, half = l&1 (8 elems).
+  // A (x-gather): CRS rows grp*16 + (l&1)*8 .. +8 of pixel m.
+  const int lm = lane >> 1;         // m offset within a 32-m slab
+  const int lh = lane & 1;          // which 8-row half of the 16-wide row
+  // per-wave A group = wv; crs row base for this lane:
+  const int acrs = rs0 + wv * 16 + lh * 8;
+  const bool aok_row = acrs < CRS;
+  const int acrsc = min(acrs, max(CRS - 8, 0));
+  int ars = acrsc / g.C;
+  const int ac = acrsc - ars * g.C;
+  const int ar = ars / g.S;
+  const int as_ = ars - ar * g.S;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    acc[i][0] = f32x4{0, 0, 0, 0};
+    acc[i][1] = f32x4{0, 0, 0, 0};
+  }
+
+  auto stage = [&](int buf, int m0) {
+    bf16_t* base = &lds[buf * 2 * W2_IMG];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {  // two 32-m slabs per 64-m tile
+      int m = m0 + i * 32 + lm;
+      bool m_ok = m < me;
+      int mcl = min(m, me - 1);
+      int n, ho = 0, wo = 0;
+      if (!flat) {
+        n = (int)mdiv40((unsigned)mcl, g.m_howo);
+        int rem = mcl - n * HoWo;
+        ho = (int)mdiv40((unsigned)rem, g.m_wo);
+        wo = rem - ho * g.Wo;
+      } else {
+        n = mcl;
+      }
+      int hi = ho * g.stride - g.pad + ar;
+      int wi = wo * g.stride - g.pad2 + as_;
+      bool ok = aok_row & m_ok & ((unsigned)hi < (unsigned)g.H)
+                & ((unsigned)wi < (unsigned)g.W);
+      const bf16_t* srcA = ok
+          ? &x[(((long)n * g.H + hi) * g.W + wi) * g.C + ac]
+          : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[wv * (WBK * 16) + i * 32 * 16], 16, 0, 0);
+      // B: dy[m][ko0 + wv*16 + lh*8 .. +8]
+      bool bok = m_ok & (ko0 + wv * 16 + lh * 8 + 8 <= g.K);
+      const bf16_t* srcB = bok
+          ? &dy[(long)mcl * g.K + ko0 + wv * 16 + lh * 8]
+          : zpage;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)
+              &base[W2_IMG + wv * (WBK * 16) + i * 32 * 16], 16, 0, 0);
+    }
+  };
+
+  const int fh = lane & 15, fq = lane >> 4;
+  const unsigned lds0 = (unsigned)(uintptr_t)&lds[0];
+  const unsigned tr_e = (unsigned)((fq * 8 + ((lane >> 2) & 3)) * 16
+                                   + (lane & 3) * 4);
+
+  const int nk = cdiv_i(me - ms, WBK);
+  if (nk > 0) stage(0, ms);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  for (int kt = 0; kt < nk; ++kt) {
+    int cur = kt & 1;
+    if (kt + 1 < nk) stage(cur ^ 1, ms + (kt + 1) * WBK);
+    const unsigned abase = lds0 + 2u * (cur * 2 * W2_IMG);
+    const unsigned bbase = abase + 2u * W2_IMG;
+    typedef __attribute__((ext_vector_type(4))) short s4_t;
+#pragma unroll
+    for (int kh = 0; kh < 2; ++kh) {
+      const unsigned khe = 2u * (tr_e + kh * 32 * 16);
+      bf16x8 afr[2], bfr[2];
+#pragma unroll
+      for (int mr = 0; mr < 2; ++mr) {
+        unsigned a = abase + 2u * ((wr * 2 + mr) * (WBK * 16)) + khe;
+        s4_t lo, hi;
+        asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                     "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+                     : "=&v"(lo), "=&v"(hi) : "v"(a));
+        afr[mr] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int nr = 0; nr < 2; ++nr) {
+        unsigned a = bbase + 2u * ((wc * 2 + nr) * (WBK * 16)) + khe;
+        s4_t lo, hi;
+        asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                     "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+                     : "=&v"(lo), "=&v"(hi) : "v"(a));
+        bfr[nr] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      if (kh == 0) {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+        __builtin_amdgcn_s_setprio(1);
+      } else {
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+      }
+#pragma unroll
+      for (int mr = 0; mr < 2; ++mr)
+#pragma unroll
+        for (int nr = 0; nr < 2; ++nr)
+          acc[mr][nr] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mr], bfr[nr], acc[mr][nr], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const long slab = (long)blockIdx.z * CRS * g.K;
+#pragma unroll
+  for (int mr = 0; mr < 2; ++mr)
+#pragma unroll
+    for (int nr = 0; nr < 2; ++nr)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = rs0 + wr * 32 + mr * 16 + fq * 4 + r;
+        int col = ko0 + wc * 32 + nr * 16 + fh;
+        if (row < CRS && col < g.K)
+          ws[slab + (long)row * g.K + col] = acc[mr][nr][r];
+      }
+}
+
 // ============================== launchers ==============================
 
 #include <stdlib.h>
@@ -1642,6 +1803,20 @@ void cilfw_conv2d_bwd_weight(const void* dy, const void* x, const void* mt,
   slice_len = cdiv(slice_len, WBK) * WBK;
   int fast_a = (C % 8 == 0);
   dim3 grid(cdiv(CRS, WBM), cdiv(K, BN), nslices);
+  void* zp = (fast_a && K % 8 == 0 && accum == 0 && conv_v2_enabled())
+                 ? zpage_ptr() : nullptr;
+  if (zp != nullptr) {  // all-glds 2-phase kernel (v2)
+    hipLaunchKernelGGL(conv2d_bwd_weight_v2_kernel, grid, dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const bf16_t*)x, (float*)ws, (const bf16_t*)zp, g,
+                       M, CRS, slice_len);
+    long len2 = (long)CRS * K;
+    hipLaunchKernelGGL(reduce_slabs_f32_kernel,
+                       dim3((int)cdiv((long)len2, (long)NTHREADS * 4)),
+                       dim3(NTHREADS), 0, (hipStream_t)stream, (float*)ws,
+                       (float*)dw, nslices, len2, accum);
+    return;
+  }
   if (fast_a)
     hipLaunchKernelGGL(conv2d_bwd_weight_kernel<true>, grid, dim3(NTHREADS),
                        0, (hipStream_t)stream, (const bf16_t*)dy,
