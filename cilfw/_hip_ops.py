@@ -34,13 +34,13 @@ _lib.cilfw_error_string.argtypes = [c_i]
 # Explicit prototypes: ctypes cannot catch arity/type mismatches on its own and
 # a wrong call corrupts device memory (see gap_bwd incident, round 1).
 _PROTOS = {
-    "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp, c_vp],
     "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 13 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
     "cilfw_im2col_smallc": [c_vp] * 3 + [c_i] * 11 + [c_vp],
     "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
-    "cilfw_bn_fwd": [c_vp] * 10 + [c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
+    "cilfw_bn_fwd": [c_vp] * 11 + [c_i, c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
     "cilfw_bn_bwd": [c_vp] * 11 + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
@@ -124,7 +124,11 @@ def _stem_cols(x, stride, pad, R, S, CRSpad):
     return col, Ho, Wo
 
 
-def conv2d_fwd(x, w, stride, pad):
+def conv2d_fwd(x, w, stride, pad, want_bn_parts=False):
+    """want_bn_parts: also produce per-M-block per-channel (sum, sumsq)
+    partials of y in the conv epilogue — a following training-mode BN skips
+    its own bn_sums pass (and a full re-read of y). Returns (y, parts|None);
+    parts is None when the shape routed off the v2 kernel or used split-K."""
     _bf16(x, "conv2d_fwd.x")
     _bf16(w, "conv2d_fwd.w")
     N, H, W_, C = x.shape
@@ -150,19 +154,24 @@ def conv2d_fwd(x, w, stride, pad):
         _lib.cilfw_conv2d_fwd(_ptr(col), _ptr(wpad), _ptr(y), _ptr(ws),
                               c_i(M), c_i(1), c_i(1), c_i(CRSpad), c_i(K),
                               c_i(1), c_i(1), c_i(1), c_i(0), c_i(1), c_i(1),
-                              c_i(ks), _stream())
+                              c_i(ks), _ptr(None), _stream())
         _check("conv2d_fwd_stem")
-        return y.view(N, Ho, Wo, K)
+        return y.view(N, Ho, Wo, K), None
     y = torch.empty(N, Ho, Wo, K, dtype=torch.bfloat16, device=x.device)
     ks = _lib.cilfw_conv2d_fwd_ksplit(N, C, K, R, S, Ho, Wo)
     ws = (torch.empty(ks * N * Ho * Wo * K, dtype=torch.float32,
                       device=x.device) if ks > 1 else None)
+    parts = None
+    if want_bn_parts and ks == 1 and C % 8 == 0             and os.environ.get("CILFW_CONV_V2", "1") != "0"             and os.environ.get("CILFW_CONV_BN_FUSE", "1") != "0":
+        gy = (N * Ho * Wo + 127) // 128
+        parts = torch.empty(gy * 2 * K, dtype=torch.float32,
+                            device=x.device)
     _lib.cilfw_conv2d_fwd(_ptr(x), _ptr(w), _ptr(y), _ptr(ws), c_i(N),
                           c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R), c_i(S),
                           c_i(stride), c_i(pad), c_i(Ho), c_i(Wo), c_i(ks),
-                          _stream())
+                          _ptr(parts), _stream())
     _check("conv2d_fwd")
-    return y
+    return y, parts
 
 
 def conv2d_bwd_data(dy, w, stride, pad, H, W_):
@@ -280,7 +289,9 @@ def conv2d_bwd_weight(dy, x, stride, pad, R, S, out=None, accum=False):
 # -------------------------------------------------------------------------- bn
 
 def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
-           relu, residual=None):
+           relu, residual=None, ext_parts=None, ext_gy=0):
+    """ext_parts/ext_gy: per-M-block (sum, sumsq) partials produced by the
+    conv that made x (conv2d_fwd want_bn_parts) — skips the bn_sums pass."""
     _bf16(x, "bn_fwd.x")
     if residual is not None:
         _bf16(residual, "bn_fwd.residual")
@@ -310,9 +321,12 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
                           device=x.device)
     gf = gamma.float().contiguous()
     bf = beta.float().contiguous()
+    if not training:
+        ext_parts = None
     _lib.cilfw_bn_fwd(_ptr(x), _ptr(y), _ptr(residual), _ptr(gf), _ptr(bf),
                       _ptr(running_mean), _ptr(running_var), _ptr(mean),
-                      _ptr(invstd), _ptr(scratch), c_l(M), c_i(C),
+                      _ptr(invstd), _ptr(scratch), _ptr(ext_parts),
+                      c_i(ext_gy), c_l(M), c_i(C),
                       c_f(momentum), c_f(eps), c_i(1 if training else 0),
                       c_i(1 if relu else 0), _stream())
     _check("bn_fwd")

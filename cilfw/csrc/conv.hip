@@ -966,7 +966,8 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
                           const bf16_t* __restrict__ w,
                           bf16_t* __restrict__ y, float* __restrict__ ws,
                           const bf16_t* __restrict__ zpage, ConvGeom g,
-                          int M, int CRS, int nk, int ksplit) {
+                          int M, int CRS, int nk, int ksplit,
+                          float* __restrict__ bn_parts) {
   constexpr int NGRP = BNX2 / 16;       // 16-col groups in the B tile
   constexpr int NRC = BNX2 / 32;        // N-fragments per wave
   constexpr int BEL = NGRP * V2BK * 16; // B image elems
@@ -1122,6 +1123,47 @@ void conv2d_fwd_v2_kernel(const bf16_t* __restrict__ x,
             y[(long)row * g.K + col] = f2bf(acc[mr][nr][r]);
         }
       }
+
+  // fused BN partial sums (a following training BN consumes these instead
+  // of re-reading y in its own bn_sums pass): per-column sum / sum-of-
+  // squares over this block's VALID rows, deterministic fixed-order
+  // reduce through the (reused) staging LDS, one [2][K] row per M-block.
+  if (bn_parts != nullptr) {
+    __syncthreads();
+    float* red = (float*)lds;  // [4 wv][4 fq][BNX2][2] = 16 KB max
+#pragma unroll
+    for (int nr = 0; nr < NRC; ++nr) {
+      float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+      for (int mr = 0; mr < 4; ++mr)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = m0 + wr * 64 + mr * 16 + fq * 4 + r;
+          // match the stored activation: the BN input is the bf16 y
+          float v = (row < M) ? bf2f(f2bf(acc[mr][nr][r])) : 0.f;
+          s1 += v;
+          s2 += v * v;
+        }
+      int cl = wc * (16 * NRC) + nr * 16 + fh;
+      red[((wv * 4 + fq) * BNX2 + cl) * 2] = s1;
+      red[((wv * 4 + fq) * BNX2 + cl) * 2 + 1] = s2;
+    }
+    __syncthreads();
+    // 2*BNX2 (col, quantity) entries; sum the 8 slots that own the col
+    for (int e = t; e < 2 * BNX2; e += NTHREADS) {
+      int cl = e >> 1, qi = e & 1;
+      int wcc = cl / (16 * NRC);
+      float a = 0.f;
+#pragma unroll
+      for (int wrr = 0; wrr < 2; ++wrr)
+#pragma unroll
+        for (int q2 = 0; q2 < 4; ++q2)
+          a += red[(((wrr * 2 + wcc) * 4 + q2) * BNX2 + cl) * 2 + qi];
+      int col = n0 + cl;
+      if (col < g.K)
+        bn_parts[((long)blockIdx.x * 2 + qi) * g.K + col] = a;
+    }
+  }
 }
 
 
@@ -1535,7 +1577,7 @@ static int fwd_bm256_min_m() {
 void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
                       int N, int H, int W, int C, int K, int R, int S,
                       int stride, int pad, int Ho, int Wo, int ksplit,
-                      void* stream) {
+                      void* bn_parts, void* stream) {
   ConvGeom g{N, H, W, C, K, R, S, stride, pad, pad, Ho, Wo,
              0, 0, magic40(C), magic40(S)};
   int M = N * Ho * Wo;
@@ -1568,7 +1610,8 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
       hipLaunchKernelGGL(conv2d_fwd_v2_kernel<128>, grid2, dim3(NTHREADS),
                          0, (hipStream_t)stream, (const bf16_t*)x,
                          (const bf16_t*)w, (bf16_t*)y, (float*)ws,
-                         (const bf16_t*)zp, g, M, CRS, nk2, ks2);
+                         (const bf16_t*)zp, g, M, CRS, nk2, ks2,
+                         (float*)(ks2 > 1 ? nullptr : bn_parts));
       if (ks2 > 1) {
         long len = (long)M * K;
         hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
@@ -1582,7 +1625,8 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
     hipLaunchKernelGGL(conv2d_fwd_v2_kernel<BN>, grid2, dim3(NTHREADS), 0,
                        (hipStream_t)stream, (const bf16_t*)x,
                        (const bf16_t*)w, (bf16_t*)y, (float*)ws,
-                       (const bf16_t*)zp, g, M, CRS, nk2, ks2);
+                       (const bf16_t*)zp, g, M, CRS, nk2, ks2,
+                       (float*)(ks2 > 1 ? nullptr : bn_parts));
     if (ks2 > 1) {
       long len = (long)M * K;
       hipLaunchKernelGGL(reduce_slabs_bf16_kernel,

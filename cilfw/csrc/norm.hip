@@ -561,11 +561,21 @@ extern "C" {
 void cilfw_bn_fwd(const void* x, void* y, const void* res,
                   const void* gamma, const void* beta,
                   void* running_mean, void* running_var, void* mean,
-                  void* invstd, void* scratch_sums, long M, int C,
+                  void* invstd, void* scratch_sums, const void* ext_part,
+                  int ext_gy, long M, int C,
                   float momentum, float eps, int training, int relu,
                   void* stream) {
   hipStream_t st = (hipStream_t)stream;
-  if (training) {
+  if (training && ext_part != nullptr) {
+    // partial sums already produced by the PRODUCING conv's epilogue
+    // (conv2d_fwd_v2 bn_parts) — one fused reduce+finalize launch replaces
+    // the bn_sums pass (and its full re-read of x)
+    hipLaunchKernelGGL(bn_reduce_finalize_kernel,
+                       dim3(cdiv(C, NT / WAVE)), dim3(NT), 0, st,
+                       (const float*)ext_part, (float*)mean, (float*)invstd,
+                       (float*)running_mean, (float*)running_var, ext_gy, M,
+                       C, momentum, eps);
+  } else if (training) {
     // scratch_sums: [gy][2][C] partials + [2][C] reduced (see wrapper sizing)
     int rows_per_blk = 256;
     dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));

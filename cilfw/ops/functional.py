@@ -76,7 +76,14 @@ class Conv2dNHWC(torch.autograd.Function):
         ctx.w_ref = wp  # grad-sink lookup (engine flat-slot delivery)
         ctx.sinked = w_param is not None
         if use_hip(x):
-            return ext().conv2d_fwd(x, wc, stride, padding)
+            y, parts = ext().conv2d_fwd(x, wc, stride, padding,
+                                        want_bn_parts=ctx.sinked)
+            if parts is not None:
+                # a following training-mode BN consumes these instead of
+                # running its own bn_sums pass over y
+                y._cilfw_bn_parts = (parts, (y.numel() // y.shape[-1]
+                                             + 127) // 128)
+            return y
         # CPU reference: fp32 NCHW conv
         xf = _to_nchw(x).float()
         wf = w.permute(3, 2, 0, 1).contiguous().float()  # (K,C,R,S)
@@ -211,9 +218,12 @@ class BatchNormAct(torch.autograd.Function):
         ctx.b_ref = b_param if b_param is not None else beta
         ctx.sinked = g_param is not None
         if use_hip(x):
+            pg = getattr(x, "_cilfw_bn_parts", None)
             y, save_mean, save_invstd = ext().bn_fwd(
                 x, gamma, beta, running_mean, running_var,
-                momentum, eps, training, relu)
+                momentum, eps, training, relu,
+                ext_parts=pg[0] if pg else None,
+                ext_gy=pg[1] if pg else 0)
             ctx.training = training
             ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
             return y
@@ -305,9 +315,12 @@ class BatchNormAddReLU(torch.autograd.Function):
         ctx.b_ref = b_param if b_param is not None else beta
         ctx.sinked = g_param is not None
         if use_hip(x):
+            pg = getattr(x, "_cilfw_bn_parts", None)
             y, save_mean, save_invstd = ext().bn_fwd(
                 x, gamma, beta, running_mean, running_var, momentum, eps,
-                training, True, residual=residual)
+                training, True, residual=residual,
+                ext_parts=pg[0] if pg else None,
+                ext_gy=pg[1] if pg else 0)
             ctx.training = training
             ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
             return y
