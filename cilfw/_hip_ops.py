@@ -54,9 +54,9 @@ _PROTOS = {
     "cilfw_stride_scatter": [c_vp] * 2 + [c_i] * 7 + [c_vp],
     "cilfw_maxpool_fwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
     "cilfw_maxpool_bwd": [c_vp] * 3 + [c_i] * 9 + [c_vp],
-    "cilfw_linear_fwd": [c_vp] * 4 + [c_i] * 3 + [c_vp],
-    "cilfw_linear_dx": [c_vp] * 3 + [c_i] * 3 + [c_vp],
-    "cilfw_linear_dw": [c_vp] * 4 + [c_i] * 3 + [c_vp],
+    "cilfw_linear_fwd": [c_vp] * 5 + [c_i] * 4 + [c_vp],
+    "cilfw_linear_dx": [c_vp] * 4 + [c_i] * 4 + [c_vp],
+    "cilfw_linear_dw": [c_vp] * 5 + [c_i] * 4 + [c_vp],
     "cilfw_ce_fwd": [c_vp] * 5 + [c_i, c_i, c_f, c_vp],
     "cilfw_ce_bwd": [c_vp] * 4 + [c_i, c_i, c_f, c_vp],
     "cilfw_kd_fwd": [c_vp] * 6 + [c_i, c_i, c_f, c_vp],
@@ -73,7 +73,8 @@ for _name, _args in _PROTOS.items():
     _fn.argtypes = _args
     _fn.restype = None
 
-for _name, _n in [("cilfw_conv2d_fwd_ksplit", 7),
+for _name, _n in [("cilfw_linear_ksplit", 3),
+                  ("cilfw_conv2d_fwd_ksplit", 7),
                   ("cilfw_conv2d_bwd_data_ksplit", 7),
                   ("cilfw_conv2d_bwd_weight_nslices", 7)]:
     _fn = getattr(_lib, _name)
@@ -449,8 +450,11 @@ def linear_fwd(x, w, bias):
     N, _ = w.shape
     y = torch.empty(M, N, dtype=torch.bfloat16, device=x.device)
     bf = bias.float().contiguous() if bias is not None else None
-    _lib.cilfw_linear_fwd(_ptr(x), _ptr(w), _ptr(bf), _ptr(y), c_i(M), c_i(N),
-                          c_i(K), _stream())
+    ks = _lib.cilfw_linear_ksplit(M, N, K)
+    ws = (torch.empty(ks * M * N, dtype=torch.float32, device=x.device)
+          if ks > 1 else None)
+    _lib.cilfw_linear_fwd(_ptr(x), _ptr(w), _ptr(bf), _ptr(y), _ptr(ws),
+                          c_i(M), c_i(N), c_i(K), c_i(ks), _stream())
     _check("linear_fwd")
     return y
 
@@ -463,10 +467,16 @@ def linear_bwd(dy, x, w, has_bias):
     dw = torch.empty(N, K, dtype=torch.float32, device=x.device)
     db = torch.empty(N, dtype=torch.float32, device=x.device) if has_bias \
         else None
-    _lib.cilfw_linear_dx(_ptr(dy), _ptr(w), _ptr(dx), c_i(M), c_i(N), c_i(K),
-                         _stream())
-    _lib.cilfw_linear_dw(_ptr(dy), _ptr(x), _ptr(dw), _ptr(db), c_i(M),
-                         c_i(N), c_i(K), _stream())
+    ks_dx = _lib.cilfw_linear_ksplit(M, K, N)
+    ws_dx = (torch.empty(ks_dx * M * K, dtype=torch.float32, device=x.device)
+             if ks_dx > 1 else None)
+    _lib.cilfw_linear_dx(_ptr(dy), _ptr(w), _ptr(dx), _ptr(ws_dx), c_i(M),
+                         c_i(N), c_i(K), c_i(ks_dx), _stream())
+    ks_dw = _lib.cilfw_linear_ksplit(N, K, M)
+    ws_dw = (torch.empty(ks_dw * N * K, dtype=torch.float32, device=x.device)
+             if ks_dw > 1 else None)
+    _lib.cilfw_linear_dw(_ptr(dy), _ptr(x), _ptr(dw), _ptr(db), _ptr(ws_dw),
+                         c_i(M), c_i(N), c_i(K), c_i(ks_dw), _stream())
     _check("linear_bwd")
     return dx, dw, db
 

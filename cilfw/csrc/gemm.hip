@@ -21,12 +21,19 @@
 #define GBS_OFF(buf) (2 * GBM * GLP + (buf) * GBN * GLP)
 #define GLDS_ELEMS (2 * GBM * GLP + 2 * GBN * GLP)
 
+// slab reduces live in conv.hip (same .so)
+__global__ void reduce_slabs_bf16_kernel(const float*, bf16_t*, int, long);
+__global__ void reduce_slabs_f32_kernel(const float*, float*, int, long, int);
+
 // MODE: 0 = NT (fwd), 1 = NN (dx), 2 = TN (dw)
+// ksplit: the head shapes are tiny in M,N but long in K (64 x 10..1000 x
+// 2048 -> ONE 64x64 block, 166 us measured on rn50) — blockIdx.z slices
+// the K loop into fp32 slabs reduced by the conv slab kernels.
 template <int MODE, bool OUT_F32>
 __global__ __launch_bounds__(GNT)
 void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
                  void* __restrict__ Cout, const float* __restrict__ bias,
-                 int M, int N, int K) {
+                 int M, int N, int K, float* __restrict__ ws, int ksplit) {
   __shared__ bf16_t lds[GLDS_ELEMS];
   const int m0 = blockIdx.x * GBM;
   const int n0 = blockIdx.y * GBN;
@@ -39,11 +46,19 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0, 0, 0, 0};
 
-  const int nk = cdiv_i(K, GBK);
+  // reduction length per mode (MODE1 reduces N, MODE2 reduces M): the
+  // round-1 kernel looped cdiv(K) steps for every mode — on the head
+  // backward that was 64 K-steps with only 1-2 carrying data (rest masked
+  // to zero), most of the measured 40-166 us
+  const int red = (MODE == 0) ? K : (MODE == 1 ? N : M);
+  const int nk = cdiv_i(red, GBK);
+  const int steps = (nk + ksplit - 1) / ksplit;
+  const int kt0 = blockIdx.z * steps;
+  const int kt1 = min(kt0 + steps, nk);
   // staging: thread t loads 8 elems of At and 8 of Bt: row = t&63, kk = t>>6+4i
   const int srow = t & 63, sk0 = t >> 6;
 
-  for (int kt = 0; kt < nk; ++kt) {
+  for (int kt = kt0; kt < kt1; ++kt) {
     const int k0 = kt * GBK;
     bf16_t* At = &lds[GAS_OFF(kt & 1)];
     bf16_t* Bt = &lds[GBS_OFF(kt & 1)];
@@ -99,8 +114,12 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
         int col = n0 + wc * 32 + nr * 16 + fi.half;
         if (row < outM && col < outN) {
           float v = acc[mr][nr][r];
-          if (MODE == 0 && bias != nullptr) v += bias[col];
-          if (OUT_F32)
+          // bias folds into slice 0's partial under split-K (deterministic)
+          if (MODE == 0 && bias != nullptr && blockIdx.z == 0)
+            v += bias[col];
+          if (ksplit > 1)
+            ws[((long)blockIdx.z * outM + row) * outN + col] = v;
+          else if (OUT_F32)
             ((float*)Cout)[(long)row * outN + col] = v;
           else
             ((bf16_t*)Cout)[(long)row * outN + col] = f2bf(v);
@@ -121,28 +140,60 @@ void colsum_kernel(const bf16_t* __restrict__ dy, float* __restrict__ db,
 
 extern "C" {
 
-void cilfw_linear_fwd(const void* x, const void* w, const void* bias, void* y,
-                      int M, int N, int K, void* stream) {
-  dim3 grid(cdiv(M, GBM), cdiv(N, GBN));
-  hipLaunchKernelGGL((gemm_kernel<0, false>), grid, dim3(GNT), 0,
-                     (hipStream_t)stream, (const bf16_t*)x, (const bf16_t*)w,
-                     y, (const float*)bias, M, N, K);
+int cilfw_linear_ksplit(int rows, int cols, int red) {
+  // fill ~256 CUs: tiny head tiles get K-sliced (>= 4 K-steps per slice)
+  int blocks = cdiv(rows, GBM) * cdiv(cols, GBN);
+  int nk = cdiv(red, GBK);
+  int ks = 1;
+  while (ks < 32 && blocks * ks < 256 && nk / (ks * 2) >= 4) ks *= 2;
+  return ks;
 }
 
-void cilfw_linear_dx(const void* dy, const void* w, void* dx, int M, int N,
-                     int K, void* stream) {
-  dim3 grid(cdiv(M, GBM), cdiv(K, GBN));
+void cilfw_linear_fwd(const void* x, const void* w, const void* bias, void* y,
+                      void* ws, int M, int N, int K, int ksplit,
+                      void* stream) {
+  dim3 grid(cdiv(M, GBM), cdiv(N, GBN), ksplit);
+  hipLaunchKernelGGL((gemm_kernel<0, false>), grid, dim3(GNT), 0,
+                     (hipStream_t)stream, (const bf16_t*)x, (const bf16_t*)w,
+                     y, (const float*)bias, M, N, K, (float*)ws, ksplit);
+  if (ksplit > 1) {
+    long len = (long)M * N;
+    hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                       dim3((int)cdiv((long)len, 1024L)), dim3(256), 0,
+                       (hipStream_t)stream, (float*)ws, (bf16_t*)y, ksplit,
+                       len);
+  }
+}
+
+void cilfw_linear_dx(const void* dy, const void* w, void* dx, void* ws,
+                     int M, int N, int K, int ksplit, void* stream) {
+  dim3 grid(cdiv(M, GBM), cdiv(K, GBN), ksplit);
   hipLaunchKernelGGL((gemm_kernel<1, false>), grid, dim3(GNT), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)w,
-                     dx, nullptr, M, N, K);
+                     dx, nullptr, M, N, K, (float*)ws, ksplit);
+  if (ksplit > 1) {
+    long len = (long)M * K;
+    hipLaunchKernelGGL(reduce_slabs_bf16_kernel,
+                       dim3((int)cdiv((long)len, 1024L)), dim3(256), 0,
+                       (hipStream_t)stream, (float*)ws, (bf16_t*)dx, ksplit,
+                       len);
+  }
 }
 
 void cilfw_linear_dw(const void* dy, const void* x, void* dw, void* db,
-                     int M, int N, int K, void* stream) {
-  dim3 grid(cdiv(N, GBM), cdiv(K, GBN));
+                     void* ws, int M, int N, int K, int ksplit,
+                     void* stream) {
+  dim3 grid(cdiv(N, GBM), cdiv(K, GBN), ksplit);
   hipLaunchKernelGGL((gemm_kernel<2, true>), grid, dim3(GNT), 0,
                      (hipStream_t)stream, (const bf16_t*)dy, (const bf16_t*)x,
-                     dw, nullptr, M, N, K);
+                     dw, nullptr, M, N, K, (float*)ws, ksplit);
+  if (ksplit > 1) {
+    long len = (long)N * K;
+    hipLaunchKernelGGL(reduce_slabs_f32_kernel,
+                       dim3((int)cdiv((long)len, 1024L)), dim3(256), 0,
+                       (hipStream_t)stream, (float*)ws, (float*)dw, ksplit,
+                       len, 0);
+  }
   if (db != nullptr)
     hipLaunchKernelGGL(colsum_kernel, dim3(cdiv(N, 256)), dim3(256), 0,
                        (hipStream_t)stream, (const bf16_t*)dy, (float*)db, M,
