@@ -516,6 +516,50 @@ void maxpool_bwd_kernel(const bf16_t* __restrict__ dy,
   dx[i] = f2bf(acc);
 }
 
+// 8-channel vectorized variant (C % 8 == 0): the scalar form moved ~2 B per
+// thread and measured 280 us on the rn50 stem pool backward
+__global__ __launch_bounds__(NT)
+void maxpool_bwd_v_kernel(const bf16_t* __restrict__ dy,
+                          const int* __restrict__ idx,
+                          bf16_t* __restrict__ dx,
+                          int N, int H, int W, int C, int kk, int st,
+                          int pad, int Ho, int Wo, long total8) {
+  long t8 = (long)blockIdx.x * NT + threadIdx.x;
+  if (t8 >= total8) return;
+  long i0 = t8 * 8;
+  int c0 = (int)(i0 % C);
+  long rest = i0 / C;
+  int wi = (int)(rest % W);
+  rest /= W;
+  int hi = (int)(rest % H);
+  int n = (int)(rest / H);
+  int my = hi * W + wi;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  int ho_lo = max(0, (hi + pad - kk + st) / st);
+  int ho_hi = min(Ho - 1, (hi + pad) / st);
+  int wo_lo = max(0, (wi + pad - kk + st) / st);
+  int wo_hi = min(Wo - 1, (wi + pad) / st);
+  for (int ho = ho_lo; ho <= ho_hi; ++ho)
+    for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+      long o = (((long)n * Ho + ho) * Wo + wo) * C + c0;
+      int4 iv0 = *(const int4*)&idx[o];
+      int4 iv1 = *(const int4*)&idx[o + 4];
+      int4 dv = *(const int4*)&dy[o];
+      const int* ie0 = (const int*)&iv0;
+      const int* ie1 = (const int*)&iv1;
+      const bf16_t* de = (const bf16_t*)&dv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (ie0[j] == my) acc[j] += bf2f(de[j]);
+        if (ie1[j] == my) acc[4 + j] += bf2f(de[4 + j]);
+      }
+    }
+  bf16_t out[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = f2bf(acc[j]);
+  *(int4*)&dx[i0] = *(int4*)out;
+}
+
 // ---------------------- strided 1x1 helpers (gather/scatter subsampled grid)
 
 __global__ __launch_bounds__(NT)
@@ -531,6 +575,46 @@ void stride_gather_kernel(const bf16_t* __restrict__ x,
   int ho = (int)(rest % Ho);
   int n = (int)(rest / Ho);
   xg[i] = x[(((long)n * H + ho * s) * W + wo * s) * C + c];
+}
+
+// 8-channel vectorized gather/scatter (C % 8 == 0)
+__global__ __launch_bounds__(NT)
+void stride_gather_v_kernel(const bf16_t* __restrict__ x,
+                            bf16_t* __restrict__ xg, int H, int W, int C,
+                            int s, int Ho, int Wo, long total8) {
+  long t8 = (long)blockIdx.x * NT + threadIdx.x;
+  if (t8 >= total8) return;
+  long i0 = t8 * 8;
+  int c = (int)(i0 % C);
+  long rest = i0 / C;
+  int wo = (int)(rest % Wo);
+  rest /= Wo;
+  int ho = (int)(rest % Ho);
+  int n = (int)(rest / Ho);
+  *(int4*)&xg[i0] =
+      *(const int4*)&x[(((long)n * H + ho * s) * W + wo * s) * C + c];
+}
+
+__global__ __launch_bounds__(NT)
+void stride_scatter_v_kernel(const bf16_t* __restrict__ dxs,
+                             bf16_t* __restrict__ dx, int H, int W, int C,
+                             int s, int Ho, int Wo, long total8) {
+  long t8 = (long)blockIdx.x * NT + threadIdx.x;
+  if (t8 >= total8) return;
+  long i0 = t8 * 8;
+  int c = (int)(i0 % C);
+  long rest = i0 / C;
+  int w = (int)(rest % W);
+  rest /= W;
+  int h = (int)(rest % H);
+  int n = (int)(rest / H);
+  int4 v{0, 0, 0, 0};
+  if (h % s == 0 && w % s == 0) {
+    int ho = h / s, wo = w / s;
+    if (ho < Ho && wo < Wo)
+      v = *(const int4*)&dxs[(((long)n * Ho + ho) * Wo + wo) * C + c];
+  }
+  *(int4*)&dx[i0] = v;
 }
 
 __global__ __launch_bounds__(NT)
@@ -713,28 +797,54 @@ void cilfw_maxpool_bwd(const void* dy, const void* idx, void* dx, int N,
                        int H, int W, int C, int kk, int st, int pad, int Ho,
                        int Wo, void* stream) {
   long total = (long)N * H * W * C;
-  hipLaunchKernelGGL(maxpool_bwd_kernel,
-                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
-                     (hipStream_t)stream, (const bf16_t*)dy, (const int*)idx,
-                     (bf16_t*)dx, N, H, W, C, kk, st, pad, Ho, Wo, total);
+  if (C % 8 == 0) {
+    long t8 = total / 8;
+    hipLaunchKernelGGL(maxpool_bwd_v_kernel,
+                       dim3((int)cdiv((long)t8, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const int*)idx, (bf16_t*)dx, N, H, W, C, kk, st,
+                       pad, Ho, Wo, t8);
+  } else {
+    hipLaunchKernelGGL(maxpool_bwd_kernel,
+                       dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)dy,
+                       (const int*)idx, (bf16_t*)dx, N, H, W, C, kk, st,
+                       pad, Ho, Wo, total);
+  }
 }
 
 void cilfw_stride_gather(const void* x, void* xg, int N, int H, int W,
                          int C, int s, int Ho, int Wo, void* stream) {
   long total = (long)N * Ho * Wo * C;
-  hipLaunchKernelGGL(stride_gather_kernel,
-                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
-                     (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)xg, H, W,
-                     C, s, Ho, Wo, total);
+  if (C % 8 == 0) {
+    long t8 = total / 8;
+    hipLaunchKernelGGL(stride_gather_v_kernel,
+                       dim3((int)cdiv((long)t8, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)xg,
+                       H, W, C, s, Ho, Wo, t8);
+  } else {
+    hipLaunchKernelGGL(stride_gather_kernel,
+                       dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)x, (bf16_t*)xg,
+                       H, W, C, s, Ho, Wo, total);
+  }
 }
 
 void cilfw_stride_scatter(const void* dxs, void* dx, int N, int H, int W,
                           int C, int s, int Ho, int Wo, void* stream) {
   long total = (long)N * H * W * C;
-  hipLaunchKernelGGL(stride_scatter_kernel,
-                     dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
-                     (hipStream_t)stream, (const bf16_t*)dxs, (bf16_t*)dx, H,
-                     W, C, s, Ho, Wo, total);
+  if (C % 8 == 0) {
+    long t8 = total / 8;
+    hipLaunchKernelGGL(stride_scatter_v_kernel,
+                       dim3((int)cdiv((long)t8, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)dxs, (bf16_t*)dx,
+                       H, W, C, s, Ho, Wo, t8);
+  } else {
+    hipLaunchKernelGGL(stride_scatter_kernel,
+                       dim3((int)cdiv((long)total, (long)NT)), dim3(NT), 0,
+                       (hipStream_t)stream, (const bf16_t*)dxs, (bf16_t*)dx,
+                       H, W, C, s, Ho, Wo, total);
+  }
 }
 
 }  // extern "C"
