@@ -1904,7 +1904,11 @@ int cilfw_conv2d_bwd_weight_nslices(int N, int C, int K, int R, int S,
   int M = N * Ho * Wo;
   int CRS = C * R * S;
   int nblocks = cdiv(CRS, WBM) * cdiv(K, BN);
-  int target = 512;
+  static int target = -1;
+  if (target < 0) {
+    const char* e = getenv("CILFW_CONV_DW_TARGET");
+    target = e ? atoi(e) : 1024;  // swept 256-2048: 1024 best (41.3k bench)
+  }
   int ns = cdiv(target, max(nblocks, 1));
   int max_ns = max(cdiv(M, WBK * 8), 1);  // keep >= 8 K-steps per slice
   if (ns > max_ns) ns = max_ns;
