@@ -1517,9 +1517,14 @@ static int bk64_max_blocks() {
 }
 
 static int pick_ksplit(int nblocks, int nk) {
-  // aim for >= 512 workgroups (2 per CU) without shredding the K loop
+  // aim for >= target workgroups without shredding the K loop
+  static int target = -1;
+  if (target < 0) {
+    const char* e = getenv("CILFW_CONV_KSPLIT_TARGET");
+    target = e ? atoi(e) : 512;
+  }
   int ks = 1;
-  while (ks < 8 && nblocks * ks < 512 && nk / (ks * 2) >= 4) ks *= 2;
+  while (ks < 8 && nblocks * ks < target && nk / (ks * 2) >= 4) ks *= 2;
   return ks;
 }
 
