@@ -35,13 +35,13 @@ _lib.cilfw_error_string.argtypes = [c_i]
 # a wrong call corrupts device memory (see gap_bwd incident, round 1).
 _PROTOS = {
     "cilfw_conv2d_fwd": [c_vp] * 4 + [c_i] * 12 + [c_vp, c_vp],
-    "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp],
+    "cilfw_conv2d_bwd_data": [c_vp] * 4 + [c_i] * 12 + [c_vp] * 5 + [c_i, c_vp],
     "cilfw_conv2d_bwd_weight": [c_vp] * 5 + [c_i] * 13 + [c_vp],
     "cilfw_fill_mtable": [c_vp] + [c_i] * 4 + [c_vp],
     "cilfw_im2col_smallc": [c_vp] * 3 + [c_i] * 11 + [c_vp],
     "cilfw_bn_apply_only": [c_vp] * 7 + [c_l, c_i, c_i, c_vp],
     "cilfw_bn_fwd": [c_vp] * 11 + [c_i, c_l, c_i, c_f, c_f, c_i, c_i, c_vp],
-    "cilfw_bn_bwd": [c_vp] * 11 + [c_l, c_i, c_i, c_i, c_vp],
+    "cilfw_bn_bwd": [c_vp] * 11 + [c_vp, c_i] + [c_l, c_i, c_i, c_i, c_vp],
     "cilfw_add_relu_fwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_add_relu_bwd": [c_vp] * 3 + [c_l, c_vp],
     "cilfw_downsample_a_fwd": [c_vp] * 2 + [c_i] * 4 + [c_vp],
@@ -73,9 +73,13 @@ for _name, _args in _PROTOS.items():
     _fn.argtypes = _args
     _fn.restype = None
 
+_lib.cilfw_conv2d_bwd_data.restype = c_i  # 1 = BN partials were emitted
+
 for _name, _n in [("cilfw_linear_ksplit", 3),
                   ("cilfw_conv2d_fwd_ksplit", 7),
                   ("cilfw_conv2d_bwd_data_ksplit", 7),
+                  ("cilfw_conv2d_bwd_data_can_fuse_bn", 8),
+                  ("cilfw_conv2d_bwd_data_bn_gy", 3),
                   ("cilfw_conv2d_bwd_weight_nslices", 7)]:
     _fn = getattr(_lib, _name)
     _fn.argtypes = [c_i] * _n
@@ -175,7 +179,12 @@ def conv2d_fwd(x, w, stride, pad, want_bn_parts=False):
     return y, parts
 
 
-def conv2d_bwd_data(dy, w, stride, pad, H, W_):
+def conv2d_bwd_data(dy, w, stride, pad, H, W_, bn_meta=None):
+    """bn_meta = (bn_y, bn_x, bn_mean, bn_invstd, bn_relu): this conv's input
+    is a training BN(+ReLU) output with a single consumer — the kernel
+    epilogue then also emits that BN's backward (dgamma, dbeta) partials so
+    the BN can skip its sums pass. Returns (dx, parts-or-None) when bn_meta
+    is given, plain dx otherwise."""
     _bf16(dy, "conv2d_bwd_data.dy")
     N, Ho, Wo, K = dy.shape
     R, S, C, Kw = w.shape
@@ -198,11 +207,33 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_):
     ks = _lib.cilfw_conv2d_bwd_data_ksplit(N, H, W_, C, K, R, S)
     ws = (torch.empty(ks * N * H * W_ * C, dtype=torch.float32,
                       device=dy.device) if ks > 1 else None)
-    _lib.cilfw_conv2d_bwd_data(_ptr(dy), _ptr(w), _ptr(dx), _ptr(ws), c_i(N),
-                               c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R),
-                               c_i(S), c_i(stride), c_i(pad), c_i(Ho),
-                               c_i(Wo), c_i(ks), _stream())
+    parts = None
+    if (bn_meta is not None
+            and _lib.cilfw_conv2d_bwd_data_can_fuse_bn(
+                N, H, W_, C, K, R, S, stride)):
+        gy = _lib.cilfw_conv2d_bwd_data_bn_gy(N, H, W_)
+        parts = torch.empty(gy, 2, C, dtype=torch.float32, device=dy.device)
+    if parts is not None:
+        bn_y, bn_x, bn_mean, bn_invstd, bn_relu = bn_meta
+        fused = _lib.cilfw_conv2d_bwd_data(
+            _ptr(dy), _ptr(w), _ptr(dx), _ptr(ws), c_i(N),
+            c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R),
+            c_i(S), c_i(stride), c_i(pad), c_i(Ho),
+            c_i(Wo), c_i(ks), _ptr(bn_y), _ptr(bn_x), _ptr(bn_mean),
+            _ptr(bn_invstd), _ptr(parts), c_i(1 if bn_relu else 0),
+            _stream())
+        if not fused:
+            parts = None
+    else:
+        _lib.cilfw_conv2d_bwd_data(
+            _ptr(dy), _ptr(w), _ptr(dx), _ptr(ws), c_i(N),
+            c_i(H), c_i(W_), c_i(C), c_i(K), c_i(R),
+            c_i(S), c_i(stride), c_i(pad), c_i(Ho),
+            c_i(Wo), c_i(ks), c_vp(0), c_vp(0), c_vp(0), c_vp(0), c_vp(0),
+            c_i(0), _stream())
     _check("conv2d_bwd_data")
+    if bn_meta is not None:
+        return dx, parts
     return dx
 
 
@@ -335,25 +366,35 @@ def bn_fwd(x, gamma, beta, running_mean, running_var, momentum, eps, training,
 
 
 def bn_bwd(dy, x, gamma, mean, invstd, y, relu, training, want_dres=False,
-           out_gamma=None, out_beta=None):
+           out_gamma=None, out_beta=None, ext_parts=None):
     """With out_gamma/out_beta the reduced per-channel grads are written
     straight into the given fp32 buffers (flat-grad slot delivery) and the
-    returned dgamma/dbeta are those buffers."""
+    returned dgamma/dbeta are those buffers. ext_parts: [gy][2][C]
+    (dgamma, dbeta) partials already emitted by the producing conv's
+    bwd-data epilogue — the sums pass over (dy, x, y) is skipped."""
     _bf16(dy, "bn_bwd.dy")
     C = x.shape[-1]
     M = x.numel() // C
     dx = torch.empty_like(x)
     dres = torch.empty_like(x) if want_dres else None
-    gy = (M + 255) // 256  # keep in sync with rows_per_blk=256 in norm.hip
-    dgb = torch.empty((gy + 1) * 2 * C, dtype=torch.float32, device=x.device)
+    if ext_parts is not None:
+        ext_gy = ext_parts.shape[0]
+        dgb = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+        base = 0
+    else:
+        ext_gy = 0
+        gy = (M + 255) // 256  # keep in sync with rows_per_blk in norm.hip
+        dgb = torch.empty((gy + 1) * 2 * C, dtype=torch.float32,
+                          device=x.device)
+        base = gy * 2 * C
     gf = gamma.float().contiguous()
     _lib.cilfw_bn_bwd(_ptr(dy), _ptr(x), _ptr(y), _ptr(dx), _ptr(dres),
                       _ptr(gf), _ptr(mean), _ptr(invstd), _ptr(dgb),
                       _ptr(out_gamma), _ptr(out_beta),
+                      _ptr(ext_parts), c_i(ext_gy),
                       c_l(M), c_i(C), c_i(1 if relu else 0),
                       c_i(1 if training else 0), _stream())
     _check("bn_bwd")
-    base = gy * 2 * C
     dgamma = out_gamma if out_gamma is not None else dgb[base:base + C]
     dbeta = out_beta if out_beta is not None else dgb[base + C:base + 2 * C]
     return dx, dgamma, dbeta, dres

@@ -708,22 +708,29 @@ void cilfw_bn_apply_only(const void* x, void* y, const void* res,
 void cilfw_bn_bwd(const void* dy, const void* x, const void* y, void* dx,
                   void* dres, const void* gamma, const void* mean,
                   const void* invstd, void* dgb, void* dg_out, void* db_out,
+                  const void* ext_part, int ext_gy,
                   long M, int C, int relu, int training, void* stream) {
   // dgb: [gy][2][C] partials followed by the reduced [dgamma | dbeta];
-  // dg_out/db_out (optional) divert the reduced grads into flat-grad slots
+  // dg_out/db_out (optional) divert the reduced grads into flat-grad slots.
+  // ext_part/ext_gy: [gy][2][C] (dgamma, dbeta) partials ALREADY produced by
+  // the upstream conv's bwd-data epilogue (this dy is that conv's dx) — the
+  // sums pass and its 3-stream re-read of (dy, x, y) are skipped; dgb then
+  // only needs room for the reduced [dgamma | dbeta].
   hipStream_t st = (hipStream_t)stream;
   int rows_per_blk = 256;
   dim3 grid(cdiv(C, 64), cdiv((int)min(M, (long)INT32_MAX), rows_per_blk));
-  float* part = (float*)dgb;
-  float* dgamma = dg_out ? (float*)dg_out : part + (long)grid.y * 2 * C;
-  float* dbeta = db_out ? (float*)db_out
-                        : part + (long)grid.y * 2 * C + C;
-  hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
-                     (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
-                     (const float*)mean, (const float*)invstd, part,
-                     M, C, rows_per_blk, relu);
+  int gy = ext_part ? ext_gy : (int)grid.y;
+  long red_off = ext_part ? 0 : (long)grid.y * 2 * C;
+  float* part = ext_part ? (float*)ext_part : (float*)dgb;
+  float* dgamma = dg_out ? (float*)dg_out : (float*)dgb + red_off;
+  float* dbeta = db_out ? (float*)db_out : (float*)dgb + red_off + C;
+  if (ext_part == nullptr)
+    hipLaunchKernelGGL(bn_bwd_sums_kernel, grid, dim3(NT), 0, st,
+                       (const bf16_t*)dy, (const bf16_t*)x, (const bf16_t*)y,
+                       (const float*)mean, (const float*)invstd, part,
+                       M, C, rows_per_blk, relu);
   hipLaunchKernelGGL(bn_reduce_slabs_kernel, dim3(cdiv(2 * C, NT / WAVE)),
-                     dim3(NT), 0, st, part, dgamma, (int)grid.y,
+                     dim3(NT), 0, st, part, dgamma, gy,
                      (long)2 * C, dbeta, (long)C);
   long total = M * C;
   long blocks = cdiv((long)total, (long)NT * 8);

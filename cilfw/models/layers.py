@@ -34,10 +34,17 @@ class BatchNormAct2d(nn.Module):
     """BN over NHWC with optional fused ReLU. gamma=1, beta=0 init
     (reference resnet.py:88-90)."""
 
-    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False):
+    def __init__(self, num_features, eps=1e-5, momentum=0.1, relu=False,
+                 fuse_bwd=True):
         super().__init__()
         self.num_features, self.eps, self.momentum, self.relu = (
             num_features, eps, momentum, relu)
+        # fuse_bwd=False: this BN's output has MULTIPLE consumers (e.g. a
+        # stem feeding both a conv and an identity skip) — the downstream
+        # conv's bwd-data epilogue must not waste work emitting BN-backward
+        # partials that can never be consumed (the accumulated grad tensor
+        # loses the attachment). Single-consumer block BNs keep the default.
+        self.fuse_bwd = fuse_bwd
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
@@ -60,7 +67,8 @@ class BatchNormAct2d(nn.Module):
                                          self.training)
         return CF.batchnorm_act(x, self.weight, self.bias, self.running_mean,
                                 self.running_var, self.momentum, self.eps,
-                                self.training, self.relu)
+                                self.training, self.relu,
+                                fuse_bwd=self.fuse_bwd)
 
     def extra_repr(self):
         return f"{self.num_features}, relu={self.relu}"

@@ -67,7 +67,10 @@ class ResNet(nn.Module):
             self.stem = Conv2d(in_channels, 64, 3, stride=1, padding=1)
         else:
             self.stem = Conv2d(in_channels, 64, 7, stride=2, padding=3)
-        self.stem_bn = BatchNormAct2d(64, relu=True)
+        # small_input: stem BN output feeds layer1's first conv AND its
+        # identity skip — two consumers, so conv-side BN-backward fusion
+        # could never be consumed (with a maxpool stem it is simply unused)
+        self.stem_bn = BatchNormAct2d(64, relu=True, fuse_bwd=False)
         self.in_ch = 64
         self.layer1 = self._make_layer(block, 64, layers[0], stride=1)
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)

@@ -46,7 +46,8 @@ class CifarResNet(nn.Module):
         n = (depth - 2) // 6
         self.depth = depth
         self.conv_1_3x3 = Conv2d(in_channels, 16, 3, stride=1, padding=1)
-        self.bn_1 = BatchNormAct2d(16, relu=True)
+        self.bn_1 = BatchNormAct2d(16, relu=True, fuse_bwd=False)  # feeds
+        # both stage_1's first conv AND its identity skip
         self.stage_1 = self._make_stage(16, 16, n, stride=1)
         self.stage_2 = self._make_stage(16, 32, n, stride=2)
         self.stage_3 = self._make_stage(32, 64, n, stride=2)

@@ -50,6 +50,34 @@ def _sink_on():
     return os.environ.get("CILFW_SINK_MODE", "join") != "off"
 
 
+def _bnbwd_fuse_on():
+    # default off: measured slower than the standalone vectorized sums pass
+    # on both rn18 and rn50 (see conv.hip bnbwd_fuse_enabled) — the kernel
+    # path stays covered by tests either way and flips on via env
+    import os
+    return os.environ.get("CILFW_BNBWD_FUSE", "0") == "1"
+
+
+def _conv_bwd_data_with_bn(ctx, E, dy, wc, stride, padding, x):
+    """bwd-data; when this conv's input is a single-consumer training-mode
+    BN(+ReLU) output, the bwd-data kernel epilogue also emits that BN's
+    backward (dgamma, dbeta) partials — attached to the returned dx, which
+    IS the BN's incoming grad, so its backward skips the sums pass. The
+    save_mean tensor rides along as an identity tag: the BN only accepts
+    partials stamped with its own per-call statistics."""
+    meta = getattr(ctx, "bnbwd_src", None)
+    if meta is None or not _bnbwd_fuse_on():
+        return E.conv2d_bwd_data(dy, wc, stride, padding,
+                                 x.shape[1], x.shape[2])
+    bn_x, bn_mean, bn_invstd, bn_relu = meta
+    dx, parts = E.conv2d_bwd_data(
+        dy, wc, stride, padding, x.shape[1], x.shape[2],
+        bn_meta=(x, bn_x, bn_mean, bn_invstd, bn_relu))
+    if parts is not None:
+        dx._cilfw_bnbwd = (parts, bn_mean)
+    return dx
+
+
 class Conv2dNHWC(torch.autograd.Function):
     """2D convolution, NHWC activations, (R,S,C,K) weight, symmetric padding.
 
@@ -75,6 +103,10 @@ class Conv2dNHWC(torch.autograd.Function):
         ctx.w_dtype = w.dtype
         ctx.w_ref = wp  # grad-sink lookup (engine flat-slot delivery)
         ctx.sinked = w_param is not None
+        # input produced by a single-consumer training BN(+ReLU): backward
+        # can emit that BN's (dgamma, dbeta) partials from the bwd-data
+        # epilogue (tensors held as plain refs — the BN's ctx saves them too)
+        ctx.bnbwd_src = getattr(x, "_cilfw_bnbwd_src", None)
         if use_hip(x):
             y, parts = ext().conv2d_fwd(x, wc, stride, padding,
                                         want_bn_parts=ctx.sinked)
@@ -111,16 +143,15 @@ class Conv2dNHWC(torch.autograd.Function):
                 with torch.cuda.stream(ws):
                     ext().conv2d_bwd_weight(dy, x, stride, padding, R, S,
                                             out=out, accum=accum)
-                dx = ext().conv2d_bwd_data(dy, wc, stride, padding,
-                                           x.shape[1], x.shape[2]) \
+                dx = _conv_bwd_data_with_bn(ctx, ext(), dy, wc, stride, padding, x) \
                     if need_dx else None
                 cur.wait_stream(ws)
                 eng.sink_delivered(pi)
                 return dx, None, None, None, None
             with torch.cuda.stream(ws):  # dW concurrent with dX (fork/join)
                 dw = ext().conv2d_bwd_weight(dy, x, stride, padding, R, S)
-            dx = ext().conv2d_bwd_data(dy, wc, stride, padding, x.shape[1],
-                                       x.shape[2]) if need_dx else None
+            dx = _conv_bwd_data_with_bn(ctx, ext(), dy, wc, stride, padding, x) \
+                if need_dx else None
             cur.wait_stream(ws)
             dw.record_stream(cur)
             return dx, dw.to(ctx.w_dtype), None, None, None
@@ -168,7 +199,7 @@ def _gb_sinkable(x, gamma, beta):
     return sg is not None and sb is not None and sg[0] is sb[0]
 
 
-def _gb_deliver_hip(eng, gi, bi, bn_bwd_args, want_dres):
+def _gb_deliver_hip(eng, gi, bi, bn_bwd_args, want_dres, ext_parts=None):
     """BN backward with dgamma/dbeta written into the flat slots. During
     grad accumulation the reduced grads go to scratch first (the dx formula
     needs THIS batch's dgamma/dbeta, not the accumulated slot) and are added."""
@@ -176,12 +207,14 @@ def _gb_deliver_hip(eng, gi, bi, bn_bwd_args, want_dres):
     bs, bacc = eng.sink_acquire(bi)
     if gacc or bacc:
         dx, dgamma, dbeta, dres = ext().bn_bwd(*bn_bwd_args,
-                                               want_dres=want_dres)
+                                               want_dres=want_dres,
+                                               ext_parts=ext_parts)
         gs.add_(dgamma)
         bs.add_(dbeta)
     else:
         dx, _, _, dres = ext().bn_bwd(*bn_bwd_args, want_dres=want_dres,
-                                      out_gamma=gs, out_beta=bs)
+                                      out_gamma=gs, out_beta=bs,
+                                      ext_parts=ext_parts)
     eng.sink_delivered(gi)
     eng.sink_delivered(bi)
     return dx, dres
@@ -206,7 +239,7 @@ class BatchNormAct(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, momentum, eps,
-                training, relu, g_param=None, b_param=None):
+                training, relu, g_param=None, b_param=None, fuse_bwd=True):
         ctx.relu = relu
         ctx.eps = eps
         # grad-sink: g_param/b_param set => gamma/beta are DETACHED (see
@@ -226,6 +259,13 @@ class BatchNormAct(torch.autograd.Function):
                 ext_gy=pg[1] if pg else 0)
             ctx.training = training
             ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+            if training and fuse_bwd and _bnbwd_fuse_on():
+                # a single downstream conv may emit this BN's backward
+                # partials from its bwd-data epilogue (models opt OUT for
+                # multi-consumer BN outputs, e.g. stems feeding identity
+                # skips — correctness is preserved either way, the partials
+                # would just be dead work)
+                y._cilfw_bnbwd_src = (x, save_mean, save_invstd, relu)
             return y
         N, H, W, C = x.shape
         xf = x.float().reshape(-1, C)
@@ -256,17 +296,23 @@ class BatchNormAct(torch.autograd.Function):
         sink = (getattr(ctx.g_ref, "_cilfw_sink", None)
                 if ctx.sinked else None)
         if use_hip(dy):
+            pgb = getattr(dy, "_cilfw_bnbwd", None)
+            # accept only partials stamped with THIS call's statistics (the
+            # save_mean tensor identity ties producer and consumer)
+            ext_parts = (pgb[0] if pgb is not None and pgb[1] is mean
+                         else None)
             if sink is not None:  # reduce writes straight into the slots
                 eng, gi = sink
                 bi = ctx.b_ref._cilfw_sink[1]
                 dx, _ = _gb_deliver_hip(
                     eng, gi, bi,
                     (dy, x, gamma, mean, invstd, y, relu, ctx.training),
-                    want_dres=False)
-                return (dx, None, None) + (None,) * 8
+                    want_dres=False, ext_parts=ext_parts)
+                return (dx, None, None) + (None,) * 9
             dx, dgamma, dbeta, _ = ext().bn_bwd(dy, x, gamma, mean, invstd,
-                                                y, relu, ctx.training)
-            return (dx, dgamma, dbeta) + (None,) * 8
+                                                y, relu, ctx.training,
+                                                ext_parts=ext_parts)
+            return (dx, dgamma, dbeta) + (None,) * 9
         C = x.shape[-1]
         dyf = dy.float().reshape(-1, C)
         if relu:
@@ -286,18 +332,18 @@ class BatchNormAct(torch.autograd.Function):
             eng, gi = sink
             bi = ctx.b_ref._cilfw_sink[1]
             _gb_deliver_cpu(eng, gi, bi, dgamma, dbeta)
-            return (dx, None, None) + (None,) * 8
-        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype)) + (None,) * 8
+            return (dx, None, None) + (None,) * 9
+        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype)) + (None,) * 9
 
 
 def batchnorm_act(x, gamma, beta, running_mean, running_var, momentum=0.1, eps=1e-5,
-                  training=True, relu=False):
+                  training=True, relu=False, fuse_bwd=True):
     if _gb_sinkable(x, gamma, beta):
         return BatchNormAct.apply(x, gamma.detach(), beta.detach(),
                                   running_mean, running_var, momentum, eps,
-                                  training, relu, [gamma], [beta])
+                                  training, relu, [gamma], [beta], fuse_bwd)
     return BatchNormAct.apply(x, gamma, beta, running_mean, running_var, momentum,
-                              eps, training, relu)
+                              eps, training, relu, None, None, fuse_bwd)
 
 
 class BatchNormAddReLU(torch.autograd.Function):

@@ -366,3 +366,54 @@ def test_herding_batch_matches_single():
         single = ext().herding_select(f.float().contiguous(),
                                       mu.contiguous(), 20)
         assert torch.equal(got.cpu(), single.cpu())
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_bnbwd_fuse_partials_match_sums_pass():
+    """CILFW_BNBWD_FUSE=1: the conv bwd-data epilogue's (dgamma, dbeta)
+    partials must reduce to the same BN grads as the standalone
+    bn_bwd_sums pass, and dx must stay bit-identical (default-off perf
+    knob — measured slower; kept env-gated). Subprocess because the C-side
+    gate is a per-process static."""
+    import os
+    import subprocess
+    import sys
+    import textwrap
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = textwrap.dedent("""
+        import torch
+        from cilfw import _hip_ops as H
+        torch.manual_seed(0)
+        dev = "cuda"
+        N, Hh, Ww, C, K = 64, 32, 32, 64, 64   # M=65536: v1 path, ksplit=1
+        dy = torch.randn(N, Hh, Ww, K, device=dev).bfloat16() * 0.1
+        w = torch.randn(3, 3, C, K, device=dev).bfloat16() * 0.05
+        bn_x = torch.randn(N, Hh, Ww, C, device=dev).bfloat16()
+        bn_y = torch.randn(N, Hh, Ww, C, device=dev).bfloat16()  # ~half <= 0
+        mean = torch.randn(C, device=dev)
+        invstd = torch.rand(C, device=dev) + 0.5
+        gamma = torch.randn(C, device=dev)
+        dx0 = H.conv2d_bwd_data(dy, w, 1, 1, Hh, Ww)
+        dx1, parts = H.conv2d_bwd_data(dy, w, 1, 1, Hh, Ww,
+                                       bn_meta=(bn_y, bn_x, mean, invstd, 1))
+        assert parts is not None, "fusion did not engage"
+        assert torch.equal(dx0, dx1), "dx changed under the BN epilogue"
+        # reference: standalone sums pass over the SAME (dy=dx, x, y)
+        a = H.bn_bwd(dx0, bn_x, gamma, mean, invstd, bn_y, True, True)
+        b = H.bn_bwd(dx0, bn_x, gamma, mean, invstd, bn_y, True, True,
+                     ext_parts=parts)
+        torch.cuda.synchronize()
+        for i, name in [(1, "dgamma"), (2, "dbeta")]:
+            torch.testing.assert_close(a[i], b[i], rtol=2e-4, atol=2e-3)
+        # dx consumes the reduced grads, whose summation ORDER differs
+        # between the two partials sources -> ulp-level drift is expected
+        torch.testing.assert_close(a[0].float(), b[0].float(),
+                                   rtol=2e-2, atol=2e-2)
+        print("fuse-parity-ok")
+    """)
+    env = dict(os.environ, CILFW_BNBWD_FUSE="1")
+    p = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=280, cwd=repo)
+    assert p.returncode == 0 and "fuse-parity-ok" in p.stdout, \
+        p.stdout[-2000:] + p.stderr[-2000:]
