@@ -106,3 +106,36 @@ def test_crash_resume_completes_protocol(tmp_path):
     assert "task id = 1" in p.stdout
     last = [l for l in p.stdout.splitlines() if "acc1s" in l][-1]
     assert last.count(",") >= 1, f"expected 2 acc1s entries: {last}"
+
+
+@pytest.mark.timeout(600)
+def test_nproc_wrapping_runs_torchrun_gloo(tmp_path):
+    """--nproc 2 wraps the child in torch.distributed.run with a loopback
+    rendezvous and a fresh port; a tiny 1-task 2-rank gloo protocol must
+    complete cleanly through the supervisor."""
+    outdir = tmp_path / "ckpts"
+    child = tmp_path / "train2.py"
+    child.write_text(textwrap.dedent(f"""
+        import sys
+        sys.path.insert(0, {REPO!r})
+        from cilfw.config import parse_args
+        from cilfw.engine import run
+        args = parse_args([
+            "--data_set", "synthetic", "--backbone", "resnet20",
+            "--synthetic_classes", "10", "--num_bases", "5",
+            "--increment", "5", "--num_epochs", "1", "--batch_size", "16",
+            "--workers", "0", "--synthetic_train_size", "128",
+            "--memory_size", "20", "--eval_every_epoch", "0",
+            "--input_size", "16", "--no_aug", "--max_tasks", "1",
+            "--output_dir", {str(outdir)!r},
+        ] + sys.argv[1:])
+        run(args)
+    """))
+    p = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "run_elastic.py"),
+         "--output_dir", str(outdir), "--max_restarts", "0", "--nproc", "2",
+         "--", sys.executable, str(child)],
+        capture_output=True, text=True, timeout=540, cwd=REPO)
+    assert p.returncode == 0, p.stdout[-3000:] + p.stderr[-3000:]
+    assert "torch.distributed.run" in p.stdout
+    assert (outdir / "task_0.pth").exists()
