@@ -198,7 +198,9 @@ def conv2d_bwd_data(dy, w, stride, pad, H, W_, bn_meta=None):
                                   c_i(W_), c_i(C), c_i(stride), c_i(Ho),
                                   c_i(Wo), _stream())
         _check("stride_scatter")
-        return dx
+        # keep the return shape contract: callers that passed bn_meta unpack
+        # a (dx, parts) pair (no fusion on the strided-1x1 scatter path)
+        return (dx, None) if bn_meta is not None else dx
     # NOTE: a 4-way parity decomposition of stride-2 bwd-data was measured
     # SLOWER than the single zero-structured kernel (12 small launches vs one
     # big one) — kernels kept (cilfw_conv2d_bwd_data_sub/parity_scatter) for a
