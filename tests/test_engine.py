@@ -51,6 +51,22 @@ def test_device_replay_matches_host_replay(monkeypatch):
 
 
 @pytest.mark.timeout(900)
+def test_checkpoint_resume_rebuilds_device_mirror(tmp_path, monkeypatch):
+    """Resume under --gpu_data: the HBM-resident replay mirror does not live
+    in the checkpoint, so the engine rebuilds it from the restored host
+    memory on the first post-resume task (engine.py "resumed run: one
+    rebuild upload"). The resumed trajectory must match the uninterrupted
+    run exactly."""
+    monkeypatch.setenv("CILFW_GPU_DATA_ON_CPU", "1")
+    base = ["--gpu_data", "--output_dir", str(tmp_path)]
+    full = run(_args(base, epochs=2))
+    resumed = run(_args(base + ["--resume", str(tmp_path / "task_0.pth")],
+                        epochs=2))
+    assert len(resumed) == 2
+    assert resumed == full
+
+
+@pytest.mark.timeout(900)
 def test_checkpoint_resume_continues(tmp_path):
     base = ["--output_dir", str(tmp_path)]
     full = run(_args(base, epochs=2))
