@@ -1661,8 +1661,16 @@ void cilfw_conv2d_fwd(const void* x, const void* w, void* y, void* ws,
       bn128 = e ? atoi(e) : 1;
     }
     // BN=128 wins only where the A-gather is trivial (1x1 convs: rn50_1x1
-    // 173->218 TF); on 3x3 layers the halved occupancy loses (l2 348->318)
-    if (bn128 && K % 128 == 0 && R == 1 && S == 1 &&
+    // 173->218 TF); on the CIFAR 3x3 layers the halved occupancy loses
+    // (l2 348->318 at M=32768) — env CILFW_CONV_V2_BN128_MIN3X3 opens the
+    // tile to 3x3 layers at/above that M (experiment; default off)
+    static int min3x3 = -2;
+    if (min3x3 == -2) {
+      const char* e = getenv("CILFW_CONV_V2_BN128_MIN3X3");
+      min3x3 = e ? atoi(e) : -1;  // -1 = never
+    }
+    int shape_ok = (R == 1 && S == 1) || (min3x3 >= 0 && M >= min3x3);
+    if (bn128 && K % 128 == 0 && shape_ok &&
         cdiv(M, BM) * cdiv(K, 128) * ks2 >= 256) {
       dim3 grid2(cdiv(M, BM), cdiv(K, 128), ks2);
       hipLaunchKernelGGL(conv2d_fwd_v2_kernel<128>, grid2, dim3(NTHREADS),
